@@ -1,0 +1,223 @@
+"""GCBF+ — the headline algorithm (reference ``gcbfplus/algo/gcbf_plus.py``).
+
+Differences vs GCBF (reference lines cited):
+  - AdamW (wd=1e-3) + finite-guard (36-139)
+  - target CBF net with polyak tau=0.5 (116, 188-191, 228)
+  - safe labels from the horizon backprop of unsafe flags (160-174)
+  - action label is the CBF-QP-rectified action u_qp computed from the
+    TARGET CBF (193-211), solved by the batched ProxQP op (K11)
+  - h_dot loss stop-gradient split for unlabeled samples (398-408)
+  - act = 2*actor + u_ref (176-180)
+"""
+from __future__ import annotations
+
+import copy
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+from torch import Tensor
+from torch.func import functional_call
+
+from ..ops.qp import proxqp_solve
+from ..trainer.data import FlatBatch, Rollout
+from ..utils.graph import GraphBatch
+from .gcbf import GCBF, _hinge_safe, _hinge_unsafe
+from .utils import clip_grads_, horizon_safe_mask, polyak_, step_if_finite
+
+
+class GCBFPlus(GCBF):
+    def __init__(self, *args, horizon: int = 32, **kwargs):
+        kwargs.setdefault("loss_h_dot_coef", 0.2)
+        super().__init__(*args, **kwargs)
+        self.horizon = horizon
+        # AdamW with decoupled weight decay, replacing the parent's Adam
+        self.cbf_optim = torch.optim.AdamW(self.cbf.parameters(), lr=self.lr_cbf,
+                                           weight_decay=1e-3)
+        self.actor_optim = torch.optim.AdamW(self.actor.parameters(), lr=self.lr_actor,
+                                             weight_decay=1e-3)
+        self.cbf_tgt = copy.deepcopy(self.cbf)
+        for p in self.cbf_tgt.parameters():
+            p.requires_grad_(False)
+        self.qp_iters = 100
+        self.qp_relax_penalty = 1e3
+
+    @property
+    def config(self) -> dict:
+        c = super().config
+        c["horizon"] = self.horizon
+        return c
+
+    # ---- update ----------------------------------------------------------
+    def update(self, rollout: Rollout, step: int) -> dict:
+        env = self._env
+        g = rollout.graph_at(env)
+        b, T = rollout.rewards.shape[:2]
+        unsafe = env.unsafe_mask(g).reshape(b, T, self.n_agents)
+        safe = horizon_safe_mask(unsafe, self.horizon)
+        batch = self._sample_batch(rollout, safe, unsafe)
+        info = self._update_nets(batch)
+        polyak_(self.cbf_tgt, self.cbf, 0.5)
+        return info
+
+    def _sample_batch(self, rollout: Rollout, safe: Tensor, unsafe: Tensor) -> FlatBatch:
+        """Reference gcbf_plus.py:232-280 (memory sampled BEFORE appending)."""
+        if self.buffer.n_data > self.batch_size:
+            mem_r, mem_s, mem_u = self.buffer.sample(rollout.length, self.rng)
+            try:
+                unsafe_flat = self.unsafe_buffer.sample(
+                    rollout.length * rollout.time_horizon, self.rng
+                )
+            except ValueError:
+                unsafe_flat = self._flat_from_rollout(mem_r, mem_s, mem_u)
+            self._append_buffers(rollout, safe, unsafe)
+            flat = FlatBatch.cat(
+                [self._flat_from_rollout(mem_r, mem_s, mem_u),
+                 self._flat_from_rollout(rollout, safe, unsafe)]
+            )
+            return FlatBatch.cat([unsafe_flat, flat])
+        self._append_buffers(rollout, safe, unsafe)
+        return self._flat_from_rollout(rollout, safe, unsafe)
+
+    def _update_nets(self, batch: FlatBatch) -> dict:
+        u_qp = self._get_b_u_qp(batch, n_chunks=8)
+        batch = batch._replace(u_qp=u_qp)
+        info = {}
+        for _ in range(self.inner_epoch):
+            perm = torch.from_numpy(self.rng.permutation(batch.n)).to(batch.states.device)
+            n_mb = max(1, batch.n // self.batch_size)
+            for mb_idx in torch.chunk(perm, n_mb):
+                info = self._update_minibatch(batch[mb_idx])
+        return info
+
+    # ---- QP labels (reference :193-211, 299-352) ---------------------------
+    def _get_b_u_qp(self, batch: FlatBatch, n_chunks: int = 8) -> Tensor:
+        outs = []
+        n = batch.n
+        chunk = max(1, (n + n_chunks - 1) // n_chunks)
+        for i in range(0, n, chunk):
+            outs.append(self.get_qp_action(batch[i : i + chunk])[0])
+        return torch.cat(outs, dim=0)
+
+    def get_qp_action(self, batch_or_graph, relax_penalty: Optional[float] = None
+                      ) -> Tuple[Tensor, Tensor]:
+        """Batched CBF-QP: min ||u - u_ref||^2 + 10||r||^2 + penalty*r
+        s.t. -Lg_h u - r <= Lf_h + alpha*0.1*h, u in box, r >= 0.
+        Returns (u_opt (M, N, nu), r (M, N))."""
+        env = self._env
+        if isinstance(batch_or_graph, FlatBatch):
+            graph = batch_or_graph.graph(env)
+        else:
+            graph = batch_or_graph
+        relax = self.qp_relax_penalty if relax_penalty is None else relax_penalty
+        M, N, nu = graph.batch_size, self.n_agents, self.action_dim
+
+        h, h_x = self.cbf_and_jacobian(graph, self.cbf_tgt)  # (M,N), (M,N,N,S)
+        agent = graph.agent_states
+        f, gdyn = env.control_affine_dyn(agent)  # (M,N,S), (M,N,S,nu)
+        Lf_h = torch.einsum("mijs,mjs->mi", h_x, f)
+        Lg_h = torch.einsum("mijs,mjsu->miju", h_x, gdyn).reshape(M, N, N * nu)
+
+        u_lb, u_ub = env.action_lim()
+        dev = agent.device
+        u_lb = u_lb.to(dev).repeat(N)
+        u_ub = u_ub.to(dev).repeat(N)
+        u_ref = env.u_ref(graph).reshape(M, N * nu)
+
+        nv = N * nu + N
+        H = torch.eye(nv, device=dev).expand(M, nv, nv).clone()
+        H[:, N * nu :, N * nu :] *= 10.0
+        gvec = torch.cat([-u_ref, relax * torch.ones(M, N, device=dev)], dim=1)
+        eyeN = torch.eye(N, device=dev).expand(M, N, N)
+        C = -torch.cat([Lg_h, eyeN], dim=2)  # (M, N, nv)
+        bvec = Lf_h + self.alpha * 0.1 * h
+        l_box = torch.cat([u_lb, torch.zeros(N, device=dev)]).expand(M, nv)
+        u_box = torch.cat([u_ub, torch.full((N,), float("inf"), device=dev)]).expand(M, nv)
+
+        x = proxqp_solve(H, gvec, C, bvec, l_box, u_box, iters=self.qp_iters)
+        u_opt = x[:, : N * nu].reshape(M, N, nu)
+        r = x[:, N * nu :]
+        return u_opt, r
+
+    def cbf_and_jacobian(self, graph: GraphBatch, cbf_net) -> Tuple[Tensor, Tensor]:
+        """h (M, N) and dh_i/dx_j (M, N, N, S) w.r.t. agent states.
+
+        Fast path (gnn_layers == 1): with one message-passing round, h_i
+        depends only on edge slots of receiver i, so ONE backward pass w.r.t.
+        the dense edge features yields every per-agent gradient; the
+        edge-build jacobian (state diff + position clip) is applied
+        analytically (reference does N reverse passes via jax.jacobian,
+        gcbf_plus.py:310-317).
+
+        General path (gnn_layers > 1): N backward passes (row i of the
+        jacobian from grad of sum_m h[m, i]).
+        """
+        env = self._env
+        M, N, S = graph.batch_size, self.n_agents, graph.state_dim
+        states = graph.states.detach()
+        with torch.enable_grad():
+            if self.gnn_layers == 1:
+                e = env.edge_feats(graph, states).detach().requires_grad_(True)
+                h = cbf_net(graph, e).squeeze(-1)  # (M, N)
+                (ge,) = torch.autograd.grad(h.sum(), e)  # (M, N, D, E)
+                h_x = env.edge_grad_to_state_jac(graph, states, ge)
+                return h.detach(), h_x
+            # general fallback: one backward per receiver row
+            st = states.clone().requires_grad_(True)
+            e = env.edge_feats(graph, st)
+            h = cbf_net(graph, e).squeeze(-1)
+            rows = []
+            for i in range(N):
+                (gs,) = torch.autograd.grad(h[:, i].sum(), st, retain_graph=i < N - 1)
+                rows.append(gs[:, :N])  # only agent-state grads
+            h_x = torch.stack(rows, dim=1)  # (M, N, N, S)
+        return h.detach(), h_x
+
+    # ---- loss (reference :354-431) ----------------------------------------
+    def _loss(self, mb: FlatBatch) -> Tuple[Tensor, dict]:
+        env = self._env
+        g = mb.graph(env)
+        e = self._edge_feats(g)
+        h2 = self.cbf(g, e).squeeze(-1)  # (mb, N)
+        h = h2.reshape(-1)
+        safe_m = mb.safe.reshape(-1)
+        unsafe_m = mb.unsafe.reshape(-1)
+
+        loss_unsafe, acc_unsafe = _hinge_unsafe(h, unsafe_m, self.eps)
+        loss_safe, acc_safe = _hinge_safe(h, safe_m, self.eps)
+
+        # action = 2*actor + u_ref (the deployed policy)
+        action = 2 * self.actor(g, e) + env.u_ref(g)
+        next_g = env.forward_graph(g, action)
+        e2 = self._edge_feats(next_g)
+        h_next = self.cbf(next_g, e2).squeeze(-1).reshape(-1)
+        h_dot = (h_next - h) / env.dt
+
+        # stop-gradient branch: CBF params detached, actor path alive
+        det_params = {k: v.detach() for k, v in self.cbf.named_parameters()}
+        h_next_ng = functional_call(self.cbf, det_params, (next_g, e2)).squeeze(-1).reshape(-1)
+        h_dot_ng = (h_next_ng - h.detach()) / env.dt
+
+        labeled = safe_m | unsafe_m
+        val = torch.relu(-h_dot - self.alpha * h + self.eps)
+        val_ng = torch.relu(-h_dot_ng - self.alpha * h + self.eps)
+        loss_h_dot = torch.where(labeled, val, val_ng).mean()
+        acc_h_dot = (h_dot + self.alpha * h > 0).float().mean()
+
+        loss_action = (action - mb.u_qp).square().sum(-1).mean()
+
+        total = (
+            self.loss_action_coef * loss_action
+            + self.loss_unsafe_coef * loss_unsafe
+            + self.loss_safe_coef * loss_safe
+            + self.loss_h_dot_coef * loss_h_dot
+        )
+        with torch.no_grad():
+            info = {
+                "loss/action": float(loss_action), "loss/unsafe": float(loss_unsafe),
+                "loss/safe": float(loss_safe), "loss/h_dot": float(loss_h_dot),
+                "loss/total": float(total), "acc/unsafe": float(acc_unsafe),
+                "acc/safe": float(acc_safe), "acc/h_dot": float(acc_h_dot),
+                "acc/unsafe_data_ratio": float(unsafe_m.float().mean()),
+            }
+        return total, info

@@ -1,0 +1,120 @@
+"""Attention message-passing GNN over the dense GraphBatch layout.
+
+Computes exactly the reference GNN (``/root/reference/gcbfplus/nn/gnn.py:44-104``):
+  msg   = Dense_msg( MLP_{256,256}( [edge_feat, sender_feat, recv_feat] ) )
+  gate  = Dense_1( MLP_{128,128}( msg ) )
+  attn  = softmax over incoming edges of each receiver (masked)
+  aggr  = sum attn * msg
+  node' = Dense_out( MLP_{256,256}( [node_feat, aggr] ) )
+but on the dense (B, N, D) slot layout where receivers are always agents and
+the softmax is a masked row softmax (see utils/graph.py docstring) — no
+segment scatter ops. The final layer updates only agent nodes (the reference
+returns type_nodes(0) anyway, gnn.py:100-104).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch import Tensor, nn
+
+from .. import ops
+from ..utils.graph import GraphBatch
+from .mlp import MLP, Dense
+
+
+def sender_index(n_agents: int, n_rays: int, device) -> Tensor:
+    """(N, D) node indices of the sender in each edge slot (computable gather
+    pattern — slot d of receiver i): agents | own goal | own lidar hits."""
+    n, r = n_agents, n_rays
+    d = n + 1 + r
+    idx = torch.empty(n, d, dtype=torch.long, device=device)
+    idx[:, :n] = torch.arange(n, device=device)[None, :]
+    idx[:, n] = torch.arange(n, device=device) + n
+    base = 2 * n + torch.arange(n, device=device)[:, None] * r
+    idx[:, n + 1 :] = base + torch.arange(r, device=device)[None, :]
+    return idx
+
+
+class GNNLayer(nn.Module):
+    def __init__(self, node_dim: int, edge_dim: int, msg_dim: int, out_dim: int,
+                 hid_msg=(256, 256), hid_aggr=(128, 128), hid_update=(256, 256)):
+        super().__init__()
+        self.msg_mlp = MLP(edge_dim + 2 * node_dim, hid_msg, act="relu", act_final=False)
+        self.msg_out = Dense(hid_msg[-1], msg_dim)
+        self.attn_mlp = MLP(msg_dim, hid_aggr, act="relu", act_final=False)
+        self.attn_out = Dense(hid_aggr[-1], 1)
+        self.update_mlp = MLP(node_dim + msg_dim, hid_update, act="relu", act_final=False)
+        self.update_out = Dense(hid_update[-1], out_dim)
+        self.msg_dim = msg_dim
+        self.out_dim = out_dim
+
+    def forward(
+        self,
+        node_feats: Tensor,  # (B, V, F)
+        edge_feats: Tensor,  # (B, N, D, E)
+        mask: Tensor,  # (B, N, D)
+        send_idx: Tensor,  # (N, D)
+        agents_only: bool,
+    ) -> Tensor:
+        B, V, F = node_feats.shape
+        N, D = mask.shape[1], mask.shape[2]
+        flat_idx = send_idx.reshape(-1)  # (N*D,)
+        sender = node_feats[:, flat_idx].reshape(B, N, D, F)
+        recv = node_feats[:, :N, None, :].expand(B, N, D, F)
+        msg_in = torch.cat([edge_feats, sender, recv], dim=-1)
+        msg = self.msg_out(self.msg_mlp(msg_in))  # (B,N,D,msg_dim)
+        gate = self.attn_out(self.attn_mlp(msg)).squeeze(-1)  # (B,N,D)
+        aggr = ops.masked_softmax_aggr(gate, msg, mask)  # (B,N,msg_dim)
+        if agents_only:
+            upd_in = torch.cat([node_feats[:, :N], aggr.to(node_feats.dtype)], dim=-1)
+        else:
+            aggr_full = torch.zeros(B, V, self.msg_dim, dtype=node_feats.dtype,
+                                    device=node_feats.device)
+            aggr_full = torch.cat([aggr.to(node_feats.dtype), aggr_full[:, N:]], dim=1)
+            upd_in = torch.cat([node_feats, aggr_full], dim=-1)
+        return self.update_out(self.update_mlp(upd_in))
+
+
+class GNN(nn.Module):
+    """n_layers of GNNLayer; returns per-agent features (B, N, out_dim)."""
+
+    def __init__(self, node_dim: int, edge_dim: int, msg_dim: int = 128, out_dim: int = 128,
+                 n_layers: int = 1, hid_msg=(256, 256), hid_aggr=(128, 128),
+                 hid_update=(256, 256)):
+        super().__init__()
+        layers = []
+        d_node = node_dim
+        for i in range(n_layers):
+            od = out_dim if i == n_layers - 1 else msg_dim
+            layers.append(GNNLayer(d_node, edge_dim, msg_dim, od, hid_msg, hid_aggr, hid_update))
+            d_node = od
+        self.layers = nn.ModuleList(layers)
+        self.node_dim = node_dim
+        self.out_dim = out_dim
+
+    def forward(self, graph: GraphBatch, edge_feats: Tensor, node_feats: Optional[Tensor] = None
+                ) -> Tensor:
+        B = graph.batch_size
+        N, R, V = graph.n_agents, graph.n_rays, graph.n_nodes
+        device = edge_feats.device
+        if node_feats is None:
+            node_feats = one_hot_node_feats(B, N, R, device, edge_feats.dtype)
+        send_idx = sender_index(N, R, device)
+        n_layers = len(self.layers)
+        x = node_feats
+        for i, layer in enumerate(self.layers):
+            last = i == n_layers - 1
+            x = layer(x, edge_feats, graph.mask, send_idx, agents_only=last)
+        return x  # (B, N, out_dim)
+
+
+def one_hot_node_feats(B: int, N: int, R: int, device, dtype=torch.float32) -> Tensor:
+    """Constant node features: agent=001, goal=010, obstacle=100
+    (reference env/double_integrator.py:288-295)."""
+    V = 2 * N + N * R
+    f = torch.zeros(V, 3, device=device, dtype=dtype)
+    f[:N, 2] = 1.0
+    f[N : 2 * N, 1] = 1.0
+    f[2 * N :, 0] = 1.0
+    return f[None].expand(B, V, 3)

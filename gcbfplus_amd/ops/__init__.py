@@ -1,0 +1,183 @@
+"""Kernel-level op surface for the MI355X build.
+
+Every hot op has two backends:
+  - CPU: plain PyTorch (fp32) — the numerics oracle and the no-GPU test path.
+  - GPU (gfx950): hand-written HIP/CDNA4 kernels in gcbfplus_amd/ops/hip/,
+    loaded as the in-tree extension ``gcbfplus_amd._C``.
+
+On a CUDA/ROCm device these ops REQUIRE the extension: if it is missing we
+raise instead of silently falling back to eager (per-project rule: the HIP
+path must be the one that runs on GPU).
+
+Op inventory (kernel numbering follows SURVEY.md §2.7):
+  fused_linear      K3/K4: MFMA GEMM + bias + activation, with autograd
+  masked_softmax_aggr K3/K4: per-receiver masked softmax + weighted message sum
+  raytrace_rect     K1: 2D LiDAR fan vs rectangle set (no grad)
+  proxqp_solve      K11: batched dense QP (labels; no grad)
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+from torch import Tensor
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+ACT_NONE, ACT_RELU, ACT_TANH = 0, 1, 2
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from gcbfplus_amd import _C  # built in-tree by setup.py / __graft_entry__.build()
+
+        _EXT = _C
+    except ImportError as e:  # pragma: no cover
+        _EXT_ERR = str(e)
+    return _EXT
+
+
+def hip_available() -> bool:
+    return _load_ext() is not None
+
+
+def _require_ext():
+    ext = _load_ext()
+    if ext is None:
+        raise RuntimeError(
+            "gcbfplus_amd._C HIP extension is required on GPU but could not be "
+            f"imported: {_EXT_ERR}. Build it with `python setup.py build_ext --inplace` "
+            "(or __graft_entry__.build())."
+        )
+    return ext
+
+
+def _apply_act(y: Tensor, act: int) -> Tensor:
+    if act == ACT_RELU:
+        return torch.relu(y)
+    if act == ACT_TANH:
+        return torch.tanh(y)
+    return y
+
+
+# --------------------------------------------------------------------------
+# fused_linear: y = act(x @ w + b)
+# --------------------------------------------------------------------------
+class _FusedLinearHIP(torch.autograd.Function):
+    """GPU path. x: (M, K) bf16 (or f32, cast), w: (K, N) f32 master, b: (N,) f32.
+
+    Forward runs the hand-written MFMA GEMM (bf16 in, f32 accumulate) with the
+    bias+activation fused into the epilogue. Backward reuses the same GEMM for
+    dx = dz @ w^T and a deterministic split-M reduction kernel for
+    dw = x^T @ dz (+ db = colsum dz).
+    """
+
+    @staticmethod
+    def forward(ctx, x: Tensor, w: Tensor, b: Optional[Tensor], act: int):
+        ext = _require_ext()
+        x_bf = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
+        w_bf = w.to(torch.bfloat16)
+        bias = b if b is not None else torch.zeros(w.shape[1], device=w.device)
+        y = ext.gemm_bias_act(x_bf.contiguous(), w_bf.contiguous(), bias.contiguous(), act)
+        ctx.save_for_backward(x_bf, w_bf, y)
+        ctx.act = act
+        ctx.x_dtype = x.dtype
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: Tensor):
+        ext = _require_ext()
+        x_bf, w_bf, y = ctx.saved_tensors
+        dz = ext.act_bwd(dy.contiguous().to(torch.bfloat16), y, ctx.act)  # (M,N) bf16
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            # dx = dz @ w^T : feed the row-major GEMM with w^T (small, cheap)
+            wt = w_bf.t().contiguous()
+            dx = ext.gemm_bias_act(dz, wt, torch.zeros(wt.shape[1], device=wt.device), ACT_NONE)
+            dx = dx.to(ctx.x_dtype)
+        if ctx.needs_input_grad[1] or ctx.needs_input_grad[2]:
+            dw, db = ext.gemm_tn(x_bf, dz)  # f32 (K,N), (N,)
+        if not ctx.has_bias:
+            db = None
+        return dx, dw, db, None
+
+
+def fused_linear(x: Tensor, w: Tensor, b: Optional[Tensor], act: int = ACT_NONE) -> Tensor:
+    """act(x @ w + b). x: (..., K); w: (K, N) fp32 master weight; b: (N,) fp32.
+
+    GPU: bf16 MFMA kernel. CPU: fp32 torch (autograd oracle).
+    """
+    lead = x.shape[:-1]
+    x2 = x.reshape(-1, x.shape[-1])
+    if x2.is_cuda:
+        y = _FusedLinearHIP.apply(x2, w, b, act)
+    else:
+        y = torch.addmm(b, x2, w) if b is not None else x2 @ w
+        y = _apply_act(y, act)
+    return y.reshape(*lead, w.shape[1])
+
+
+# --------------------------------------------------------------------------
+# masked_softmax_aggr: attention aggregate over the dense edge-slot axis
+# --------------------------------------------------------------------------
+class _SoftmaxAggrHIP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate: Tensor, msg: Tensor, mask: Tensor):
+        ext = _require_ext()
+        gate_f = gate.to(torch.float32).contiguous()
+        msg_bf = msg.to(torch.bfloat16).contiguous()
+        aggr, attn = ext.softmax_aggr_fwd(gate_f, msg_bf, mask.contiguous())
+        ctx.save_for_backward(attn, msg_bf, mask)
+        ctx.gate_dtype = gate.dtype
+        ctx.msg_dtype = msg.dtype
+        return aggr
+
+    @staticmethod
+    def backward(ctx, daggr: Tensor):
+        ext = _require_ext()
+        attn, msg_bf, mask = ctx.saved_tensors
+        dgate, dmsg = ext.softmax_aggr_bwd(
+            daggr.to(torch.bfloat16).contiguous(), attn, msg_bf, mask
+        )
+        return dgate.to(ctx.gate_dtype), dmsg.to(ctx.msg_dtype), None
+
+
+def masked_softmax_aggr(gate: Tensor, msg: Tensor, mask: Tensor) -> Tensor:
+    """Per-receiver attention aggregation (dense replacement for the
+    reference's jraph.segment_softmax + segment_sum, nn/gnn.py:65-72).
+
+    gate: (B, N, D) raw attention logits
+    msg:  (B, N, D, C) messages
+    mask: (B, N, D) bool; masked slots get zero attention.
+    Returns aggr: (B, N, C) = sum_d softmax_d(gate | mask) * msg.
+    Rows with no active slot return zeros.
+    """
+    if gate.is_cuda:
+        return _SoftmaxAggrHIP.apply(gate, msg, mask)
+    neg = torch.finfo(gate.dtype).min
+    g = torch.where(mask, gate, torch.full_like(gate, neg))
+    # numerically-stable masked softmax; all-masked rows -> zeros
+    gmax = g.max(dim=-1, keepdim=True).values
+    e = torch.exp(g - gmax) * mask.to(gate.dtype)
+    denom = e.sum(dim=-1, keepdim=True).clamp_min(1e-20)
+    attn = e / denom
+    return torch.einsum("bnd,bndc->bnc", attn.to(msg.dtype), msg)
+
+
+# --------------------------------------------------------------------------
+# raytrace (K1) — no autograd
+# --------------------------------------------------------------------------
+def raytrace_rect(pos: Tensor, points: Tensor, n_rays: int, sense_range: float) -> Tensor:
+    """2D LiDAR fan: pos (B,N,2) origins, points (B,K,4,2) rectangle corners.
+    Returns hits (B,N,R,2). GPU: HIP kernel; CPU: composed torch (obstacle.py)."""
+    ext = _load_ext()
+    if pos.is_cuda:
+        _require_ext()
+        return _EXT.raytrace_rect(pos.contiguous(), points.contiguous(), n_rays, sense_range)
+    raise NotImplementedError("CPU path goes through env.get_lidar composition")

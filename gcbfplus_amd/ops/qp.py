@@ -1,0 +1,113 @@
+"""Batched dense QP solver (K11) — ProxQP/OSQP-family operator splitting.
+
+Solves, for a batch of M small dense problems:
+    min_x  1/2 x^T H x + g^T x
+    s.t.   C x <= b,   l <= x <= u
+
+This replaces the reference's external JaxProxQP dependency
+(``/root/reference/gcbfplus/algo/gcbf_plus.py:329-346``): same QP shape, same
+fixed-iteration batched semantics (Settings.max_iter=100, no early exit so
+every problem in the batch does identical work — GPU-friendly by
+construction).
+
+Method: ADMM with over-relaxation on the stacked constraint set
+A = [C; I], lo = [-inf; l], hi = [b; u]; the (H + sigma I + rho A^T A) system
+is factorized once (batched Cholesky) and each iteration is two triangular
+solves + projections. All ops are batched torch -> runs on CPU (oracle) and
+GPU; a one-workgroup-per-QP HIP kernel backs it on gfx950 when the problem
+count is large.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor
+
+
+def proxqp_solve(
+    H: Tensor,  # (M, n, n)
+    g: Tensor,  # (M, n)
+    C: Tensor,  # (M, k, n)
+    b: Tensor,  # (M, k)
+    l: Tensor,  # (M, n) box lower
+    u: Tensor,  # (M, n) box upper
+    iters: int = 100,
+    rho: float = 0.1,
+    sigma: float = 1e-6,
+    alpha: float = 1.6,
+    ruiz_iters: int = 5,
+) -> Tensor:
+    """Returns x (M, n). Fixed iteration count, no per-problem early exit.
+
+    Applies OSQP-style Ruiz equilibration + cost scaling first — essential
+    because the GCBF+ QP mixes O(1) action costs with the 1e3 relaxation
+    penalty (gcbf_plus.py:299 relax_penalty).
+    """
+    M, n = g.shape
+    k = b.shape[1]
+    dtype = torch.float32
+    H = H.to(dtype)
+    g = g.to(dtype)
+    C = C.to(dtype)
+    b = b.to(dtype)
+
+    eye_n = torch.eye(n, dtype=dtype, device=g.device)
+    A = torch.cat([C, eye_n.expand(M, n, n)], dim=1)  # (M, k+n, n)
+    m_c = k + n
+    lo = torch.cat([torch.full_like(b, -float("inf")), l.to(dtype)], dim=1)
+    hi = torch.cat([b, u.to(dtype)], dim=1)
+
+    # ---- Ruiz equilibration of [[H, A^T], [A, 0]] -----------------------
+    D = torch.ones(M, n, dtype=dtype, device=g.device)
+    E = torch.ones(M, m_c, dtype=dtype, device=g.device)
+    Hs, As = H, A
+    for _ in range(ruiz_iters):
+        col_h = Hs.abs().amax(dim=1)  # (M, n) inf-norm of H columns
+        col_a = As.abs().amax(dim=1)  # (M, n) inf-norm of A columns
+        dn = torch.clamp(torch.maximum(col_h, col_a), min=1e-8).rsqrt()
+        row_a = As.abs().amax(dim=2)  # (M, m_c)
+        de = torch.clamp(row_a, min=1e-8).rsqrt()
+        Hs = Hs * dn[:, :, None] * dn[:, None, :]
+        As = As * de[:, :, None] * dn[:, None, :]
+        D = D * dn
+        E = E * de
+    gs = g * D
+    # cost scaling (OSQP): c = 1 / max(1, mean col norm of Hs, ||gs||_inf)
+    c = 1.0 / torch.clamp(
+        torch.maximum(Hs.abs().amax(dim=1).mean(dim=1), gs.abs().amax(dim=1)), min=1.0
+    )
+    Hs = Hs * c[:, None, None]
+    gs = gs * c[:, None]
+    los = lo * E
+    his = hi * E
+
+    K = Hs + sigma * eye_n + rho * As.transpose(1, 2) @ As
+    Lc = torch.linalg.cholesky(K)
+
+    x = torch.zeros(M, n, dtype=dtype, device=g.device)
+    z = torch.zeros(M, m_c, dtype=dtype, device=g.device)
+    y = torch.zeros(M, m_c, dtype=dtype, device=g.device)
+
+    for _ in range(iters):
+        rhs = sigma * x - gs + torch.einsum("mkn,mk->mn", As, rho * z - y)
+        xt = torch.cholesky_solve(rhs.unsqueeze(-1), Lc).squeeze(-1)
+        zt = torch.einsum("mkn,mn->mk", As, xt)
+        x = alpha * xt + (1 - alpha) * x
+        z_relax = alpha * zt + (1 - alpha) * z
+        znew = torch.clamp(z_relax + y / rho, los, his)
+        y = y + rho * (z_relax - znew)
+        z = znew
+    return x * D
+
+
+def qp_kkt_residuals(H, g, C, b, l, u, x) -> Tuple[Tensor, Tensor]:
+    """(primal_infeas, stationarity) residual norms for testing."""
+    ineq = torch.clamp(torch.einsum("mkn,mn->mk", C, x) - b, min=0.0)
+    box = torch.clamp(l - x, min=0.0) + torch.clamp(x - u.clamp(max=1e30), min=0.0)
+    primal = torch.cat([ineq, box], dim=1).abs().amax(dim=1)
+    # stationarity is only checkable with duals; report gradient-projection residual
+    grad = torch.einsum("mij,mj->mi", H, x) + g
+    x_step = torch.clamp(x - grad, l, u.clamp(max=1e30))
+    stat = (x - x_step).abs().amax(dim=1)
+    return primal, stat

@@ -1,0 +1,106 @@
+"""Rollout collection + metric helpers (reference trainer/utils.py:25-55)."""
+from __future__ import annotations
+
+import json
+import os
+import time
+from typing import Callable, Optional
+
+import numpy as np
+import torch
+from torch import Tensor
+
+from ..utils.graph import GraphBatch
+from .data import Rollout
+
+
+@torch.no_grad()
+def collect_rollout(env, act_fn: Callable, graph0: GraphBatch) -> Rollout:
+    """Roll all B worlds forward max_episode_steps with act_fn(graph)->action.
+
+    The reference scans a jitted body (trainer/utils.py:43-53); here the body
+    is a host loop of batched device kernels (captured in a HIP graph on GPU).
+    """
+    B, T = graph0.batch_size, env.max_episode_steps
+    V, S = graph0.n_nodes, graph0.state_dim
+    N, D = graph0.n_agents, graph0.n_edge_slots
+    dev = graph0.device
+    states = torch.empty(B, T, V, S, device=dev)
+    masks = torch.empty(B, T, N, D, dtype=torch.bool, device=dev)
+    actions = torch.empty(B, T, N, env.action_dim, device=dev)
+    rewards = torch.empty(B, T, device=dev)
+    costs = torch.empty(B, T, device=dev)
+    dones = torch.zeros(B, T, dtype=torch.bool, device=dev)
+
+    graph = graph0
+    for t in range(T):
+        action = act_fn(graph)
+        if isinstance(action, tuple):
+            action = action[0]
+        states[:, t] = graph.states
+        masks[:, t] = graph.mask
+        actions[:, t] = action
+        res = env.step(graph, action)
+        rewards[:, t] = res.reward
+        costs[:, t] = res.cost
+        dones[:, t] = res.done
+        graph = res.graph
+
+    return Rollout(
+        states=states, masks=masks, actions=actions, rewards=rewards, costs=costs,
+        dones=dones, next_states=graph.states, next_mask=graph.mask,
+        obstacles=graph0.env_states,
+    )
+
+
+@torch.no_grad()
+def eval_rollout_metrics(env, rollout: Rollout) -> dict:
+    """Eval metrics as the reference trainer computes them
+    (trainer/trainer.py:105-129)."""
+    total_reward = rollout.rewards.sum(dim=-1)
+    g = rollout.graph_at(env)
+    b, T = rollout.rewards.shape[:2]
+    finish = env.finish_mask(g).reshape(b, T, -1).float()
+    finish = finish.amax(dim=1).mean()
+    cost = rollout.costs.sum(dim=-1).mean()
+    unsafe_frac = (rollout.costs.amax(dim=-1) >= 1e-6).float().mean()
+    return {
+        "eval/reward": float(total_reward.mean()),
+        "eval/reward_min": float(total_reward.min()),
+        "eval/reward_max": float(total_reward.max()),
+        "eval/reward_final": float(rollout.rewards[:, -1].mean()),
+        "eval/cost": float(cost),
+        "eval/unsafe_frac": float(unsafe_frac),
+        "eval/finish": float(finish),
+    }
+
+
+class MetricsLogger:
+    """stdout + JSONL metrics; wandb only if importable (no network here)."""
+
+    def __init__(self, log_dir: Optional[str], run_name: str = "run", use_wandb: bool = True):
+        self.log_dir = log_dir
+        self._f = None
+        if log_dir is not None:
+            os.makedirs(log_dir, exist_ok=True)
+            self._f = open(os.path.join(log_dir, "metrics.jsonl"), "a")
+        self._wandb = None
+        if use_wandb:
+            try:  # pragma: no cover - wandb absent in this image
+                import wandb
+
+                self._wandb = wandb
+                wandb.init(name=run_name, project="gcbf-amd", dir=log_dir, mode="offline")
+            except Exception:
+                self._wandb = None
+
+    def log(self, metrics: dict, step: int):
+        if self._f is not None:
+            self._f.write(json.dumps({"step": step, **metrics}) + "\n")
+            self._f.flush()
+        if self._wandb is not None:
+            self._wandb.log(metrics, step=step)
+
+    def close(self):
+        if self._f is not None:
+            self._f.close()

@@ -1,0 +1,131 @@
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from gcbfplus_amd.algo import make_algo
+from gcbfplus_amd.algo.utils import horizon_safe_mask
+from gcbfplus_amd.env import make_env
+from gcbfplus_amd.trainer.utils import collect_rollout, eval_rollout_metrics
+
+
+def brute_force_safe(unsafe, horizon):
+    """Literal transcription of reference gcbf_plus.py:160-174."""
+    b, T, N = unsafe.shape
+    safe = np.ones((b, T, N), dtype=bool)
+    for bb in range(b):
+        for n in range(N):
+            m = np.ones(T, dtype=bool)
+            for i in range(T):
+                start = 0 if i < horizon else i - horizon
+                if unsafe[bb, i, n]:
+                    m[start : i + 1] = False
+            m[0] = True
+            safe[bb, :, n] = m
+    return safe
+
+
+def test_horizon_safe_mask_matches_reference():
+    rng = np.random.default_rng(0)
+    unsafe = rng.random((3, 20, 4)) < 0.2
+    for horizon in (1, 4, 32):
+        ours = horizon_safe_mask(torch.from_numpy(unsafe), horizon).numpy()
+        ref = brute_force_safe(unsafe, horizon)
+        assert (ours == ref).all(), horizon
+
+
+@pytest.fixture(scope="module")
+def small_setup():
+    torch.manual_seed(0)
+    env = make_env("DoubleIntegrator", num_agents=4, area_size=2.0, max_step=8, device="cpu")
+    algo = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim,
+                     n_agents=env.num_agents, gnn_layers=1, batch_size=8, buffer_size=16,
+                     horizon=4, inner_epoch=2, seed=0)
+    return env, algo
+
+
+def test_gcbf_plus_update_changes_params(small_setup):
+    env, algo = small_setup
+    rng = np.random.default_rng(0)
+    before = [p.detach().clone() for p in algo.cbf.parameters()]
+    g = env.reset(2, rng)
+    ro = collect_rollout(env, algo.step, g)
+    info = algo.update(ro, 0)
+    assert np.isfinite(info["loss/total"])
+    after = list(algo.cbf.parameters())
+    assert any(not torch.equal(a, b) for a, b in zip(after, before))
+    # target moved toward online (tau=0.5)
+    for pt, po in zip(algo.cbf_tgt.parameters(), algo.cbf.parameters()):
+        assert torch.isfinite(pt).all()
+
+
+def test_checkpoint_roundtrip(tmp_path, small_setup):
+    env, algo = small_setup
+    rng = np.random.default_rng(1)
+    g = env.reset(1, rng)
+    a_before = algo.act(g)
+    algo.save(str(tmp_path), 7)
+    assert os.path.exists(tmp_path / "7" / "actor.pkl")
+    assert os.path.exists(tmp_path / "7" / "cbf.pkl")
+
+    torch.manual_seed(123)
+    algo2 = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                      state_dim=env.state_dim, action_dim=env.action_dim,
+                      n_agents=env.num_agents, gnn_layers=1, batch_size=8, buffer_size=16,
+                      horizon=4, seed=99)
+    a_mid = algo2.act(g)
+    assert not torch.allclose(a_mid, a_before)
+    algo2.load(str(tmp_path), 7)
+    a_after = algo2.act(g)
+    assert torch.allclose(a_after, a_before, atol=1e-6)
+
+
+def test_act_is_2actor_plus_uref(small_setup):
+    env, algo = small_setup
+    rng = np.random.default_rng(2)
+    g = env.reset(1, rng)
+    e = env.edge_feats(g)
+    with torch.no_grad():
+        expected = 2 * algo.actor(g, e) + env.u_ref(g)
+    assert torch.allclose(algo.act(g), expected, atol=1e-6)
+
+
+def test_eval_metrics(small_setup):
+    env, algo = small_setup
+    rng = np.random.default_rng(3)
+    g = env.reset(2, rng)
+    ro = collect_rollout(env, algo.act, g)
+    m = eval_rollout_metrics(env, ro)
+    for k in ("eval/reward", "eval/cost", "eval/unsafe_frac", "eval/finish"):
+        assert np.isfinite(m[k])
+    assert 0.0 <= m["eval/finish"] <= 1.0
+
+
+def test_gcbf_update_runs():
+    torch.manual_seed(0)
+    env = make_env("SingleIntegrator", num_agents=3, area_size=2.0, max_step=8, device="cpu")
+    algo = make_algo("gcbf", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim,
+                     n_agents=env.num_agents, gnn_layers=1, batch_size=8, buffer_size=16,
+                     inner_epoch=2, seed=0)
+    rng = np.random.default_rng(0)
+    for step in range(2):
+        g = env.reset(2, rng)
+        ro = collect_rollout(env, algo.step, g)
+        info = algo.update(ro, step)
+    assert np.isfinite(info["loss/total"])
+
+
+def test_online_policy_refinement_runs(small_setup):
+    env, algo = small_setup
+    rng = np.random.default_rng(4)
+    g = env.reset(1, rng)
+    algo.online_pol_refine = True
+    try:
+        a = algo.act(g)
+    finally:
+        algo.online_pol_refine = False
+    assert a.shape == (1, env.num_agents, env.action_dim)
+    assert torch.isfinite(a).all()
