@@ -14,6 +14,7 @@ import torch
 from torch import Tensor
 from torch.func import functional_call
 
+from ..parallel import dp
 from ..trainer.buffer import FlatSampleBuffer, MaskedRolloutBuffer
 from ..trainer.data import FlatBatch, Rollout
 from ..utils.graph import GraphBatch
@@ -72,7 +73,8 @@ class GCBF(MultiAgentController):
 
         self.buffer = MaskedRolloutBuffer(size=buffer_size)
         self.unsafe_buffer = FlatSampleBuffer(size=buffer_size // 2)
-        self.rng = np.random.default_rng(seed=seed + 1)
+        self.rng = np.random.default_rng(seed=seed + 1 + 7919 * dp.rank())
+        dp.broadcast_modules([self.cbf, self.actor])
 
     # ---- config / io -----------------------------------------------------
     @property
@@ -182,11 +184,11 @@ class GCBF(MultiAgentController):
         safe, unsafe = self._collect_masks(rollout)
         if self.buffer.n_data > self.batch_size:
             mem_r, mem_s, mem_u = self.buffer.sample(rollout.length // 2, self.rng)
-            try:
+            if dp.all_agree(self.unsafe_buffer.length > 0):
                 unsafe_flat = self.unsafe_buffer.sample(
                     rollout.length * rollout.time_horizon, self.rng
                 )
-            except ValueError:
+            else:
                 unsafe_flat = self._flat_from_rollout(mem_r, mem_s, mem_u)
             self._append_buffers(rollout, safe, unsafe)
             flat = FlatBatch.cat(
@@ -261,6 +263,7 @@ class GCBF(MultiAgentController):
         total.backward()
         cbf_params = [p for p in self.cbf.parameters()]
         actor_params = [p for p in self.actor.parameters()]
+        dp.allreduce_mean_grads(cbf_params + actor_params)
         cbf_norm = clip_grads_(cbf_params, self.max_grad_norm)
         actor_norm = clip_grads_(actor_params, self.max_grad_norm)
         step_if_finite(self.cbf_optim, cbf_params, cbf_norm)

@@ -20,6 +20,7 @@ from torch import Tensor
 from torch.func import functional_call
 
 from ..ops.qp import proxqp_solve
+from ..parallel import dp
 from ..trainer.data import FlatBatch, Rollout
 from ..utils.graph import GraphBatch
 from .gcbf import GCBF, _hinge_safe, _hinge_unsafe
@@ -64,11 +65,11 @@ class GCBFPlus(GCBF):
         """Reference gcbf_plus.py:232-280 (memory sampled BEFORE appending)."""
         if self.buffer.n_data > self.batch_size:
             mem_r, mem_s, mem_u = self.buffer.sample(rollout.length, self.rng)
-            try:
+            if dp.all_agree(self.unsafe_buffer.length > 0):
                 unsafe_flat = self.unsafe_buffer.sample(
                     rollout.length * rollout.time_horizon, self.rng
                 )
-            except ValueError:
+            else:
                 unsafe_flat = self._flat_from_rollout(mem_r, mem_s, mem_u)
             self._append_buffers(rollout, safe, unsafe)
             flat = FlatBatch.cat(

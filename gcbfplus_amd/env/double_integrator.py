@@ -121,7 +121,14 @@ class DoubleIntegrator(MultiAgentEnv):
 
     # ---- graph build (reference :288-320, 223-264) -------------------------
     def get_lidar_hits(self, agent_pos: Tensor, obstacles: Rectangle) -> Tensor:
-        """(B, N, 2) -> hit points (B, N, R, 2)."""
+        """(B, N, 2) -> hit points (B, N, R, 2). GPU: K1 HIP kernel."""
+        if agent_pos.is_cuda:
+            from .. import ops
+
+            return ops.raytrace_rect(
+                agent_pos.contiguous(), obstacles.points, self.n_rays,
+                self._params["comm_radius"],
+            )
         return get_lidar(agent_pos, obstacles, self.n_rays, self._params["comm_radius"])
 
     def build_mask(self, states: Tensor) -> Tensor:

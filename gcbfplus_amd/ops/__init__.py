@@ -94,7 +94,16 @@ class _FusedLinearHIP(torch.autograd.Function):
     def backward(ctx, dy: Tensor):
         ext = _require_ext()
         x_bf, w_bf, y = ctx.saved_tensors
-        dz = ext.act_bwd(dy.contiguous().to(torch.bfloat16), y, ctx.act)  # (M,N) bf16
+        if y.dtype == torch.float32:  # small-N f32 head output path
+            if ctx.act == ACT_RELU:
+                dz_f = dy * (y > 0)
+            elif ctx.act == ACT_TANH:
+                dz_f = dy * (1.0 - y * y)
+            else:
+                dz_f = dy
+            dz = dz_f.to(torch.bfloat16).contiguous()
+        else:
+            dz = ext.act_bwd(dy.contiguous().to(torch.bfloat16), y, ctx.act)  # (M,N) bf16
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # dx = dz @ w^T : feed the row-major GEMM with w^T (small, cheap)

@@ -1,0 +1,281 @@
+#include "hip/hip_runtime.h"
+// MFMA GEMM kernels for the GNN MLP stack.
+//
+// The networks are stacks of Dense layers with K, N in {16..384} and M up to
+// ~1e5 rows (batch * agents * edge-slots), bf16 inputs / f32 accumulate.
+// Three kernels:
+//   gemm_bias_act_kernel : Y = act(X @ W + b), 128x64 tile, BK=32,
+//                          B-operand staged once per block in a
+//                          fragment-blocked LDS image, A staged per K-tile.
+//   gemv_bias_act_kernel : N <= 16 head/gate outputs, wave-per-row.
+//   gemm_tn kernels      : dW = X^T @ dZ with a deterministic split-M
+//                          partial-sum pass + reduction (no atomics).
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// Y[M,N] = act(X[M,K] @ W[K,N] + bias[N]);  K % 32 == 0 (launcher pads).
+// grid: (ceil(M/128), ceil(N/64)); block: 256 threads (4 waves, 2x2).
+// LDS: B-block [K/8][64][8] (k-blocked so B-fragments are single b128 reads)
+//      + A-tile [128][40] (pad 32->40 kills b128 bank conflicts).
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void gemm_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
+                          const float* __restrict__ bias, bf16_t* __restrict__ Y,
+                          int M, int N, int K) {
+  constexpr int BM = 128, BN = 64, BK = 32, APAD = 40;
+  extern __shared__ char smem[];
+  bf16_t* sB = (bf16_t*)smem;                    // [K/8][BN][8]
+  bf16_t* sA = (bf16_t*)(smem + (K / 8) * BN * 8 * sizeof(bf16_t));  // [BM][APAD]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wm = w >> 1, wn = w & 1;  // wave tile: 64(M) x 32(N)
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  // ---- preload W block n0..n0+63 into k-blocked LDS image ----------------
+  // chunk c -> k = c>>3, col8 = (c&7)*8 ; threads cover K*8 chunks.
+  for (int c = tid; c < K * 8; c += 256) {
+    const int k = c >> 3;
+    const int co = (c & 7) * 8;
+    bf16_t v[8];
+    if (n0 + co + 7 < N) {
+      *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
+  }
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int arow = tid >> 1;            // A-stage: 2 threads per row
+  const int acol = (tid & 1) * 16;      // each stages 16 bf16 = 32 B
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    // stage A[m0..+128][k0..+32]
+    bf16_t av[16];
+    const long arow_g = (long)(m0 + arow);
+    if (arow_g < M) {
+      *(bf16x8*)av = *(const bf16x8*)(X + arow_g * K + k0 + acol);
+      *(bf16x8*)(av + 8) = *(const bf16x8*)(X + arow_g * K + k0 + acol + 8);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 16; ++i) av[i] = (bf16_t)0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < 16; i += 8) *(bf16x8*)(sA + arow * APAD + acol + i) = *(bf16x8*)(av + i);
+    __syncthreads();
+
+    // fragments + MFMA
+    bf16x8 afr[4];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+      afr[mf] = *(const bf16x8*)(sA + (wm * 64 + mf * 16 + (lane & 15)) * APAD + (lane >> 4) * 8);
+    bf16x8 bfr[2];
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      bfr[nf] = *(const bf16x8*)(sB + (((k0 >> 3) + (lane >> 4)) * BN + wn * 32 + nf * 16 + (lane & 15)) * 8);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+  }
+
+  // ---- epilogue: bias + activation + bf16 store --------------------------
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      const int col = n0 + wn * 32 + nf * 16 + (lane & 15);
+      if (col >= N) continue;
+      const float bv = bias[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = m0 + wm * 64 + mf * 16 + (lane >> 4) * 4 + r;
+        if (row < M) Y[row * N + col] = (bf16_t)apply_act(acc[mf][nf][r] + bv, ACT);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Small-N path (N <= 16): one wave per output row, W cached in LDS.
+// grid: ceil(M/4); block 256 (4 waves).
+// ---------------------------------------------------------------------------
+template <int ACT, typename OutT>
+__launch_bounds__(256) __global__
+void gemv_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
+                          const float* __restrict__ bias, OutT* __restrict__ Y,
+                          int M, int N, int K) {
+  extern __shared__ char smem[];
+  bf16_t* sW = (bf16_t*)smem;  // [K][N]
+  for (int i = threadIdx.x; i < K * N; i += 256) sW[i] = W[i];
+  __syncthreads();
+
+  const int lane = threadIdx.x & 63;
+  const long m = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (m >= M) return;
+  float p[16];
+#pragma unroll
+  for (int n = 0; n < 16; ++n) p[n] = 0.f;
+  for (int k = lane; k < K; k += WAVE) {
+    const float xv = (float)X[m * K + k];
+    for (int n = 0; n < N; ++n) p[n] += xv * (float)sW[k * N + n];
+  }
+  for (int n = 0; n < N; ++n) {
+    const float s = wave_reduce_sum(p[n]);
+    if (lane == 0) Y[m * N + n] = (OutT)apply_act(s + bias[n], ACT);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// act_bwd: dZ = dY * act'(Y), elementwise bf16.
+// ---------------------------------------------------------------------------
+__global__ void act_bwd_kernel(const bf16_t* __restrict__ dY, const bf16_t* __restrict__ Y,
+                               bf16_t* __restrict__ dZ, long n, int act) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dZ[i] = (bf16_t)((float)dY[i] * act_grad_from_out((float)Y[i], act));
+}
+
+// ---------------------------------------------------------------------------
+// dW = X^T @ dZ  (deterministic split-M).  Stage 1: partials[s][K][N].
+// grid: (ceil(K/64), ceil(N/64), S); block 256 (4 waves 2x2, wave 32x32).
+// A-operand = X^T staged transposed in LDS (scalar transpose writes),
+// B-operand = dZ staged k-blocked like the forward GEMM.
+// ---------------------------------------------------------------------------
+__launch_bounds__(256) __global__
+void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
+                            float* __restrict__ partial, int M, int N, int K, int S) {
+  constexpr int BKDIM = 64, BN = 64, BMR = 32, TPAD = 40;
+  __shared__ bf16_t sXT[BKDIM][TPAD];   // [k][m] transposed X tile
+  __shared__ bf16_t sB[BMR / 8][BN][8]; // dZ tile, m-blocked
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wk = w >> 1, wn = w & 1;  // wave tile 32(K) x 32(N)
+  const int k0 = blockIdx.x * BKDIM;
+  const int n0 = blockIdx.y * BN;
+  const int s = blockIdx.z;
+
+  // split-M range for this s (fixed boundaries -> deterministic)
+  const long m_per = ((long)M + S - 1) / S;
+  const long ms = (long)s * m_per;
+  const long me = (ms + m_per < (long)M) ? ms + m_per : (long)M;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (long m0 = ms; m0 < me; m0 += BMR) {
+    __syncthreads();
+    {  // stage X^T: thread t loads X[m0 + t>>3][k0 + (t&7)*8 ..+8]
+      const long mr = m0 + (tid >> 3);
+      const int kc = (tid & 7) * 8;
+      bf16_t v[8];
+      if (mr < me && k0 + kc + 7 < K) {
+        *(bf16x8*)v = *(const bf16x8*)(X + mr * K + k0 + kc);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          v[i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i) sXT[kc + i][tid >> 3] = v[i];
+    }
+    {  // stage dZ m-blocked: chunk c -> mrow = c>>3, col8 = (c&7)*8
+      for (int c = tid; c < BMR * 8; c += 256) {
+        const long mr = m0 + (c >> 3);
+        const int co = (c & 7) * 8;
+        bf16_t v[8];
+        if (mr < me && n0 + co + 7 < N) {
+          *(bf16x8*)v = *(const bf16x8*)(dZ + mr * N + n0 + co);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            v[i] = (mr < me && n0 + co + i < N) ? dZ[mr * N + n0 + co + i] : (bf16_t)0.f;
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) sB[(c >> 3) >> 3][co + i][(c >> 3) & 7] = v[i];
+      }
+    }
+    __syncthreads();
+
+    bf16x8 afr[2], bfr[2];
+#pragma unroll
+    for (int kf = 0; kf < 2; ++kf)
+      afr[kf] = *(const bf16x8*)(&sXT[wk * 32 + kf * 16 + (lane & 15)][(lane >> 4) * 8]);
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      bfr[nf] = *(const bf16x8*)(&sB[lane >> 4][wn * 32 + nf * 16 + (lane & 15)][0]);
+#pragma unroll
+    for (int kf = 0; kf < 2; ++kf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
+  }
+
+  float* out = partial + (long)s * K * N;
+#pragma unroll
+  for (int kf = 0; kf < 2; ++kf)
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      const int col = n0 + wn * 32 + nf * 16 + (lane & 15);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int krow = k0 + wk * 32 + kf * 16 + (lane >> 4) * 4 + r;
+        if (krow < K) out[(long)krow * N + col] = acc[kf][nf][r];
+      }
+    }
+}
+
+// partials (S,K,N) f32 -> dW (K,N) f32 (fixed-order sum: deterministic)
+__global__ void reduce_partials_kernel(const float* __restrict__ partial,
+                                       float* __restrict__ dW, long KN, int S) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= KN) return;
+  float a = 0.f;
+  for (int s = 0; s < S; ++s) a += partial[(long)s * KN + i];
+  dW[i] = a;
+}
+
+// db[n] = sum_m dZ[m][n]: split-M partials (grid.y = S) then reduce via
+// reduce_partials_kernel. Deterministic (fixed split boundaries + order).
+__global__ void colsum_partial_kernel(const bf16_t* __restrict__ dZ, float* __restrict__ partial,
+                                      long M, int N, int S) {
+  const int n = blockIdx.x * blockDim.x + threadIdx.x;
+  const int s = blockIdx.y;
+  if (n >= N) return;
+  const long m_per = (M + S - 1) / S;
+  const long ms = (long)s * m_per;
+  const long me = (ms + m_per < M) ? ms + m_per : M;
+  float a = 0.f;
+  for (long m = ms; m < me; ++m) a += (float)dZ[m * N + n];
+  partial[(long)s * N + n] = a;
+}
+
+// ---------------------------------------------------------------------------
+// host-visible launchers (called from bindings.cpp)
+// ---------------------------------------------------------------------------
+template __global__ void gemm_bias_act_kernel<0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemv_bias_act_kernel<0, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemv_bias_act_kernel<1, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemv_bias_act_kernel<2, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemv_bias_act_kernel<0, float>(const bf16_t*, const bf16_t*, const float*, float*, int, int, int);
+template __global__ void gemv_bias_act_kernel<1, float>(const bf16_t*, const bf16_t*, const float*, float*, int, int, int);
+template __global__ void gemv_bias_act_kernel<2, float>(const bf16_t*, const bf16_t*, const float*, float*, int, int, int);

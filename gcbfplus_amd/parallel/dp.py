@@ -1,0 +1,80 @@
+"""Data parallelism over RCCL/xGMI (C1 in SURVEY.md §2.7).
+
+The reference is strictly single-device; DP here shards the
+n_env_train worlds across ranks (one process per GPU, torch.distributed with
+backend "nccl" == RCCL on ROCm). Each rank runs rollout + QP labels + loss
+fwd/bwd on its own shard; gradients are averaged with ONE fused-bucket
+all-reduce per minibatch (the two nets' grads are ~2.9 MB fp32 total, so
+latency dominates on 7x153 GB/s xGMI — a single flat bucket beats many small
+calls; overlap-with-backward is pointless at this size).
+"""
+from __future__ import annotations
+
+import os
+from typing import Iterable, List
+
+import torch
+import torch.distributed as dist
+
+
+def is_active() -> bool:
+    return dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1
+
+
+def rank() -> int:
+    return dist.get_rank() if is_active() else 0
+
+
+def world_size() -> int:
+    return dist.get_world_size() if is_active() else 1
+
+
+def setup_from_env() -> int:
+    """Init process group from torchrun env vars; returns local rank."""
+    if "WORLD_SIZE" not in os.environ or int(os.environ["WORLD_SIZE"]) <= 1:
+        return 0
+    if not dist.is_initialized():
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend=backend)
+    local = int(os.environ.get("LOCAL_RANK", 0))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local)
+    return local
+
+
+def broadcast_modules(modules: Iterable[torch.nn.Module]):
+    if not is_active():
+        return
+    for m in modules:
+        for p in m.parameters():
+            dist.broadcast(p.data, src=0)
+
+
+@torch.no_grad()
+def allreduce_mean_grads(params: List[torch.nn.Parameter]):
+    """One fused flat-bucket mean all-reduce over all grads."""
+    if not is_active():
+        return
+    grads = [p.grad for p in params if p.grad is not None]
+    if not grads:
+        return
+    flat = torch.cat([g.reshape(-1) for g in grads])
+    dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+    flat /= world_size()
+    off = 0
+    for g in grads:
+        n = g.numel()
+        g.copy_(flat[off : off + n].reshape(g.shape))
+        off += n
+
+
+def all_agree(flag: bool) -> bool:
+    """Collective AND — used for control-flow that must match across ranks
+    (e.g. 'is the unsafe buffer non-empty everywhere')."""
+    if not is_active():
+        return flag
+    t = torch.tensor([1 if flag else 0], dtype=torch.int64)
+    if dist.get_backend() == "nccl":
+        t = t.to("cuda")
+    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    return bool(t.item())

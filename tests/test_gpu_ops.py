@@ -1,0 +1,177 @@
+"""GPU numerics tests: every HIP kernel vs the plain-PyTorch fp32 reference
+of the same op (tolerances sized for bf16 inputs / f32 accumulate)."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from gcbfplus_amd import ops
+
+
+def _rand(*shape, scale=1.0):
+    return (torch.randn(*shape) * scale).to("cuda")
+
+
+def ref_linear(x, w, b, act):
+    y = x.to(torch.float32) @ w.to(torch.float32) + b
+    return ops._apply_act(y, act)
+
+
+@pytest.mark.parametrize("M,K,N,act", [
+    (128, 32, 64, 0),
+    (1000, 128, 256, 1),
+    (5248, 10, 256, 1),     # first-layer concat shape (K padded inside)
+    (2048, 256, 128, 0),
+    (517, 384, 256, 1),     # ragged M
+    (2048, 256, 1, 2),      # CBF head out (gemv path)
+    (2048, 256, 2, 2),      # actor out
+    (4096, 128, 1, 0),      # attn gate
+])
+def test_gemm_bias_act_matches_fp32(M, K, N, act):
+    torch.manual_seed(0)
+    x = _rand(M, K, scale=0.5)
+    w = _rand(K, N, scale=0.2)
+    b = _rand(N, scale=0.1).float()
+    y = ops.fused_linear(x, w, b, act)
+    y_ref = ref_linear(x, w, b, act)
+    err = (y.float() - y_ref).abs()
+    scale = y_ref.abs().mean().clamp_min(1.0)
+    assert (err.mean() / scale) < 5e-3, (err.max().item(), err.mean().item())
+    assert (err.max() / scale) < 8e-2
+
+
+def test_fused_linear_backward_matches_fp32():
+    torch.manual_seed(1)
+    M, K, N = 777, 128, 256
+    x = _rand(M, K, scale=0.5).requires_grad_(True)
+    w = _rand(K, N, scale=0.2).requires_grad_(True)
+    b = _rand(N, scale=0.1).float().requires_grad_(True)
+    y = ops.fused_linear(x, w, b, ops.ACT_RELU)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    xf = x.detach().float().cpu().requires_grad_(True)
+    wf = w.detach().float().cpu().requires_grad_(True)
+    bf = b.detach().float().cpu().requires_grad_(True)
+    yf = torch.relu(xf @ wf + bf)
+    yf.backward(g.float().cpu())
+
+    for got, ref in ((x.grad, xf.grad), (w.grad, wf.grad), (b.grad, bf.grad)):
+        got = got.float().cpu()
+        denom = ref.abs().mean().clamp_min(1e-3)
+        assert ((got - ref).abs().mean() / denom) < 2e-2, (got - ref).abs().max()
+
+
+def test_gemm_tn_deterministic():
+    torch.manual_seed(2)
+    x = _rand(4096, 128).to(torch.bfloat16)
+    dz = _rand(4096, 256).to(torch.bfloat16)
+    from gcbfplus_amd import _C
+    dw1, db1 = _C.gemm_tn(x, dz)
+    dw2, db2 = _C.gemm_tn(x, dz)
+    assert torch.equal(dw1, dw2) and torch.equal(db1, db2)
+
+
+def test_softmax_aggr_matches_fp32():
+    torch.manual_seed(3)
+    B, N, D, C = 8, 8, 41, 128
+    gate = _rand(B, N, D)
+    msg = _rand(B, N, D, C, scale=0.5)
+    mask = torch.rand(B, N, D, device="cuda") < 0.5
+    mask[:, :, N] = True  # goal slot always on
+    out = ops.masked_softmax_aggr(gate, msg.to(torch.bfloat16), mask)
+    ref = ops.masked_softmax_aggr(gate.cpu(), msg.cpu(), mask.cpu())
+    err = (out.float().cpu() - ref).abs()
+    assert err.mean() < 5e-3 and err.max() < 5e-2, (err.mean(), err.max())
+
+
+def test_softmax_aggr_backward_matches_fp32():
+    torch.manual_seed(4)
+    B, N, D, C = 4, 6, 20, 128
+    gate = _rand(B, N, D).requires_grad_(True)
+    msg = _rand(B, N, D, C, scale=0.5).requires_grad_(True)
+    mask = torch.rand(B, N, D, device="cuda") < 0.6
+    mask[:, :, 0] = True
+    out = ops.masked_softmax_aggr(gate, msg, mask)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    gate_c = gate.detach().cpu().requires_grad_(True)
+    msg_c = msg.detach().cpu().requires_grad_(True)
+    ref = ops.masked_softmax_aggr(gate_c, msg_c, mask.cpu())
+    ref.backward(g.float().cpu())
+    for got, refg in ((gate.grad, gate_c.grad), (msg.grad, msg_c.grad)):
+        err = (got.float().cpu() - refg).abs()
+        denom = refg.abs().mean().clamp_min(1e-4)
+        assert (err.mean() / denom) < 3e-2, (err.mean(), err.max())
+
+
+def test_softmax_aggr_all_masked_row_zero():
+    B, N, D, C = 2, 3, 10, 128
+    gate = _rand(B, N, D)
+    msg = _rand(B, N, D, C)
+    mask = torch.zeros(B, N, D, dtype=torch.bool, device="cuda")
+    mask[:, 1:] = True
+    out = ops.masked_softmax_aggr(gate, msg, mask)
+    assert torch.isfinite(out).all()
+    assert out[:, 0].abs().max().item() == 0.0
+
+
+def test_raytrace_matches_cpu():
+    from gcbfplus_amd.env import make_env
+
+    env = make_env("DoubleIntegrator", num_agents=8, area_size=4.0, max_step=4,
+                   device="cuda")
+    rng = np.random.default_rng(0)
+    obs = env.sample_obstacles(3, rng)
+    obs_gpu = type(obs)(*[t.cuda() for t in obs])
+    pos = torch.rand(3, 8, 2) * 4.0
+    hits_gpu = env.get_lidar_hits(pos.cuda(), obs_gpu).cpu()
+    from gcbfplus_amd.env.utils import get_lidar
+
+    hits_cpu = get_lidar(pos, obs, env.n_rays, env.params["comm_radius"])
+    # in-range hits must agree to fp tolerance; no-hit beams land at huge
+    # coords (1e6 scale) where tiny angle diffs blow up absolute error
+    close = hits_cpu.abs().amax(dim=-1, keepdim=True) < 100.0
+    err = (hits_gpu - hits_cpu).abs() * close
+    assert err.max() < 1e-3, err.max()
+    assert torch.isfinite(hits_gpu).all()
+
+
+def test_full_cbf_forward_gpu_vs_cpu():
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.algo.module.cbf import CBFNet
+
+    torch.manual_seed(5)
+    env_c = make_env("DoubleIntegrator", num_agents=8, area_size=4.0, max_step=4,
+                     device="cpu")
+    g = env_c.reset(4, np.random.default_rng(1))
+    net = CBFNet(env_c.node_dim, env_c.edge_dim, 1)
+    with torch.no_grad():
+        h_cpu = net(g, env_c.edge_feats(g))
+    g_gpu = g.to("cuda")
+    net_gpu = net.to("cuda")
+    with torch.no_grad():
+        h_gpu = net_gpu(g_gpu, env_c.edge_feats(g_gpu))
+    err = (h_gpu.cpu() - h_cpu).abs()
+    assert err.max() < 0.05, err.max()  # tanh-bounded output, bf16 path
+
+
+def test_gpu_update_step_finite():
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.algo import make_algo
+    from gcbfplus_amd.trainer.utils import collect_rollout
+
+    torch.manual_seed(6)
+    env = make_env("DoubleIntegrator", num_agents=8, area_size=4.0, max_step=8,
+                   device="cuda")
+    algo = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim,
+                     n_agents=env.num_agents, gnn_layers=1, batch_size=16,
+                     buffer_size=16, horizon=4, inner_epoch=1, seed=0)
+    rng = np.random.default_rng(2)
+    g = env.reset(2, rng)
+    ro = collect_rollout(env, algo.step, g)
+    info = algo.update(ro, 0)
+    assert all(np.isfinite(v) for v in info.values()), info
