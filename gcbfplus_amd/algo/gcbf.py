@@ -14,6 +14,8 @@ import torch
 from torch import Tensor
 from torch.func import functional_call
 
+from .. import ops
+from ..ops.optim import FusedAdamW
 from ..parallel import dp
 from ..trainer.buffer import FlatSampleBuffer, MaskedRolloutBuffer
 from ..trainer.data import FlatBatch, Rollout
@@ -68,13 +70,25 @@ class GCBF(MultiAgentController):
         torch.manual_seed(seed)
         self.cbf = CBFNet(node_dim, edge_dim, gnn_layers).to(dev)
         self.actor = DeterministicPolicyNet(node_dim, edge_dim, action_dim, gnn_layers).to(dev)
-        self.cbf_optim = torch.optim.Adam(self.cbf.parameters(), lr=lr_cbf)
-        self.actor_optim = torch.optim.Adam(self.actor.parameters(), lr=lr_actor)
+        self._create_optimizers()
 
         self.buffer = MaskedRolloutBuffer(size=buffer_size)
         self.unsafe_buffer = FlatSampleBuffer(size=buffer_size // 2)
         self.rng = np.random.default_rng(seed=seed + 1 + 7919 * dp.rank())
         dp.broadcast_modules([self.cbf, self.actor])
+
+    def _use_fused_optim(self) -> bool:
+        return self._env.device.type == "cuda" and ops.hip_available()
+
+    def _create_optimizers(self):
+        """Adam (reference gcbf.py:101,118); GPU: fused flat-buffer kernel
+        with the clip+finite-guard folded in (K13)."""
+        if self._use_fused_optim():
+            self.cbf_optim = FusedAdamW(self.cbf, self.lr_cbf, 0.0, self.max_grad_norm)
+            self.actor_optim = FusedAdamW(self.actor, self.lr_actor, 0.0, self.max_grad_norm)
+        else:
+            self.cbf_optim = torch.optim.Adam(self.cbf.parameters(), lr=self.lr_cbf)
+            self.actor_optim = torch.optim.Adam(self.actor.parameters(), lr=self.lr_actor)
 
     # ---- config / io -----------------------------------------------------
     @property
@@ -201,11 +215,13 @@ class GCBF(MultiAgentController):
             batch = self._flat_from_rollout(rollout, safe, unsafe)
 
         info = {}
-        for _ in range(self.inner_epoch):
+        for ep in range(self.inner_epoch):
             perm = torch.from_numpy(self.rng.permutation(batch.n)).to(batch.states.device)
             n_mb = max(1, batch.n // self.batch_size)
-            for mb_idx in torch.chunk(perm, n_mb):
-                info = self._update_minibatch(batch[mb_idx])
+            chunks = torch.chunk(perm, n_mb)
+            for i, mb_idx in enumerate(chunks):
+                last = ep == self.inner_epoch - 1 and i == len(chunks) - 1
+                info = self._update_minibatch(batch[mb_idx], want_info=last)
         return info
 
     def _append_buffers(self, rollout: Rollout, safe: Tensor, unsafe: Tensor):
@@ -215,7 +231,7 @@ class GCBF(MultiAgentController):
         sel = row_mask.reshape(-1)
         self.unsafe_buffer.append(flat[sel])
 
-    def _loss(self, mb: FlatBatch) -> Tuple[Tensor, dict]:
+    def _loss(self, mb: FlatBatch, want_info: bool = True) -> Tuple[Tensor, dict]:
         """Reference gcbf.py:258-321 loss; action target is u_ref and the
         action fed to forward_graph is the RAW actor output."""
         env = self._env
@@ -246,30 +262,38 @@ class GCBF(MultiAgentController):
             + self.loss_safe_coef * loss_safe
             + self.loss_h_dot_coef * loss_h_dot
         )
-        with torch.no_grad():
-            info = {
-                "loss/action": float(loss_action), "loss/unsafe": float(loss_unsafe),
-                "loss/safe": float(loss_safe), "loss/h_dot": float(loss_h_dot),
-                "loss/total": float(total), "acc/unsafe": float(acc_unsafe),
-                "acc/safe": float(acc_safe), "acc/h_dot": float(acc_h_dot),
-                "acc/unsafe_data_ratio": float(unsafe_m.float().mean()),
-            }
+        info = {}
+        if want_info:
+            with torch.no_grad():
+                info = {
+                    "loss/action": float(loss_action), "loss/unsafe": float(loss_unsafe),
+                    "loss/safe": float(loss_safe), "loss/h_dot": float(loss_h_dot),
+                    "loss/total": float(total), "acc/unsafe": float(acc_unsafe),
+                    "acc/safe": float(acc_safe), "acc/h_dot": float(acc_h_dot),
+                    "acc/unsafe_data_ratio": float(unsafe_m.float().mean()),
+                }
         return total, info
 
-    def _update_minibatch(self, mb: FlatBatch) -> dict:
-        total, info = self._loss(mb)
-        self.cbf_optim.zero_grad(set_to_none=True)
-        self.actor_optim.zero_grad(set_to_none=True)
+    def _update_minibatch(self, mb: FlatBatch, want_info: bool = True) -> dict:
+        total, info = self._loss(mb, want_info)
+        self.cbf_optim.zero_grad(set_to_none=False)
+        self.actor_optim.zero_grad(set_to_none=False)
         total.backward()
-        cbf_params = [p for p in self.cbf.parameters()]
-        actor_params = [p for p in self.actor.parameters()]
-        dp.allreduce_mean_grads(cbf_params + actor_params)
-        cbf_norm = clip_grads_(cbf_params, self.max_grad_norm)
-        actor_norm = clip_grads_(actor_params, self.max_grad_norm)
-        step_if_finite(self.cbf_optim, cbf_params, cbf_norm)
-        step_if_finite(self.actor_optim, actor_params, actor_norm)
-        info["grad_norm/cbf"] = float(cbf_norm)
-        info["grad_norm/actor"] = float(actor_norm)
+        if isinstance(self.cbf_optim, FusedAdamW):
+            dp.allreduce_mean_flat([self.cbf_optim.gflat, self.actor_optim.gflat])
+            cbf_norm = self.cbf_optim.step()
+            actor_norm = self.actor_optim.step()
+        else:
+            cbf_params = [p for p in self.cbf.parameters()]
+            actor_params = [p for p in self.actor.parameters()]
+            dp.allreduce_mean_grads(cbf_params + actor_params)
+            cbf_norm = clip_grads_(cbf_params, self.max_grad_norm)
+            actor_norm = clip_grads_(actor_params, self.max_grad_norm)
+            step_if_finite(self.cbf_optim, cbf_params, cbf_norm)
+            step_if_finite(self.actor_optim, actor_params, actor_norm)
+        if want_info:
+            info["grad_norm/cbf"] = float(cbf_norm)
+            info["grad_norm/actor"] = float(actor_norm)
         return info
 
 

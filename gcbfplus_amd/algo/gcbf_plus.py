@@ -32,16 +32,24 @@ class GCBFPlus(GCBF):
         kwargs.setdefault("loss_h_dot_coef", 0.2)
         super().__init__(*args, **kwargs)
         self.horizon = horizon
-        # AdamW with decoupled weight decay, replacing the parent's Adam
-        self.cbf_optim = torch.optim.AdamW(self.cbf.parameters(), lr=self.lr_cbf,
-                                           weight_decay=1e-3)
-        self.actor_optim = torch.optim.AdamW(self.actor.parameters(), lr=self.lr_actor,
-                                             weight_decay=1e-3)
         self.cbf_tgt = copy.deepcopy(self.cbf)
         for p in self.cbf_tgt.parameters():
             p.requires_grad_(False)
         self.qp_iters = 100
         self.qp_relax_penalty = 1e3
+
+    def _create_optimizers(self):
+        """AdamW wd=1e-3 (reference gcbf_plus.py:109,127)."""
+        from ..ops.optim import FusedAdamW
+
+        if self._use_fused_optim():
+            self.cbf_optim = FusedAdamW(self.cbf, self.lr_cbf, 1e-3, self.max_grad_norm)
+            self.actor_optim = FusedAdamW(self.actor, self.lr_actor, 1e-3, self.max_grad_norm)
+        else:
+            self.cbf_optim = torch.optim.AdamW(self.cbf.parameters(), lr=self.lr_cbf,
+                                               weight_decay=1e-3)
+            self.actor_optim = torch.optim.AdamW(self.actor.parameters(), lr=self.lr_actor,
+                                                 weight_decay=1e-3)
 
     @property
     def config(self) -> dict:
@@ -84,11 +92,13 @@ class GCBFPlus(GCBF):
         u_qp = self._get_b_u_qp(batch, n_chunks=8)
         batch = batch._replace(u_qp=u_qp)
         info = {}
-        for _ in range(self.inner_epoch):
+        for ep in range(self.inner_epoch):
             perm = torch.from_numpy(self.rng.permutation(batch.n)).to(batch.states.device)
             n_mb = max(1, batch.n // self.batch_size)
-            for mb_idx in torch.chunk(perm, n_mb):
-                info = self._update_minibatch(batch[mb_idx])
+            chunks = torch.chunk(perm, n_mb)
+            for i, mb_idx in enumerate(chunks):
+                last = ep == self.inner_epoch - 1 and i == len(chunks) - 1
+                info = self._update_minibatch(batch[mb_idx], want_info=last)
         return info
 
     # ---- QP labels (reference :193-211, 299-352) ---------------------------
@@ -175,7 +185,7 @@ class GCBFPlus(GCBF):
         return h.detach(), h_x
 
     # ---- loss (reference :354-431) ----------------------------------------
-    def _loss(self, mb: FlatBatch) -> Tuple[Tensor, dict]:
+    def _loss(self, mb: FlatBatch, want_info: bool = True) -> Tuple[Tensor, dict]:
         env = self._env
         g = mb.graph(env)
         e = self._edge_feats(g)
@@ -213,12 +223,14 @@ class GCBFPlus(GCBF):
             + self.loss_safe_coef * loss_safe
             + self.loss_h_dot_coef * loss_h_dot
         )
-        with torch.no_grad():
-            info = {
-                "loss/action": float(loss_action), "loss/unsafe": float(loss_unsafe),
-                "loss/safe": float(loss_safe), "loss/h_dot": float(loss_h_dot),
-                "loss/total": float(total), "acc/unsafe": float(acc_unsafe),
-                "acc/safe": float(acc_safe), "acc/h_dot": float(acc_h_dot),
-                "acc/unsafe_data_ratio": float(unsafe_m.float().mean()),
-            }
+        info = {}
+        if want_info:
+            with torch.no_grad():
+                info = {
+                    "loss/action": float(loss_action), "loss/unsafe": float(loss_unsafe),
+                    "loss/safe": float(loss_safe), "loss/h_dot": float(loss_h_dot),
+                    "loss/total": float(total), "acc/unsafe": float(acc_unsafe),
+                    "acc/safe": float(acc_safe), "acc/h_dot": float(acc_h_dot),
+                    "acc/unsafe_data_ratio": float(unsafe_m.float().mean()),
+                }
         return total, info

@@ -45,9 +45,11 @@ def step_if_finite(optim: torch.optim.Optimizer, params: List[Tensor], norm: Ten
 @torch.no_grad()
 def polyak_(target: nn.Module, online: nn.Module, tau: float):
     """target <- tau * online + (1 - tau) * target (optax.incremental_update,
-    gcbf_plus.py:188-191)."""
-    for pt, po in zip(target.parameters(), online.parameters()):
-        pt.mul_(1 - tau).add_(po, alpha=tau)
+    gcbf_plus.py:188-191). Fused multi-tensor (two kernel launches)."""
+    pts = list(target.parameters())
+    pos = list(online.parameters())
+    torch._foreach_mul_(pts, 1 - tau)
+    torch._foreach_add_(pts, pos, alpha=tau)
     for bt, bo in zip(target.buffers(), online.buffers()):
         bt.copy_(bo)
 

@@ -17,6 +17,14 @@ __global__ void colsum_partial_kernel(const bf16_t_*, float*, long, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
+__global__ void grad_norm_sq_partial_kernel(const float*, long, float*);
+__global__ void reduce_norm_kernel(const float*, int, float*);
+__global__ void adamw_flat_kernel(float*, const float*, float*, float*, const float*, const int*,
+                                  float, float, float, float, float, float, long);
+__global__ void advance_step_kernel(int*, const float*);
+template <int MAXNV, int MAXK>
+__global__ void proxqp_kernel(const float*, const float*, const float*, const float*, const float*,
+                              const float*, float*, int, int, int, int, float, float, float);
 
 #define CHECK_IN(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 
@@ -173,7 +181,66 @@ torch::Tensor raytrace_rect(torch::Tensor pos, torch::Tensor points, long n_rays
   return hits;
 }
 
+torch::Tensor fused_adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                               torch::Tensor v, torch::Tensor t, double lr, double b1,
+                               double b2, double eps, double wd, double max_norm) {
+  CHECK_IN(p);
+  CHECK_IN(g);
+  CHECK_IN(m);
+  CHECK_IN(v);
+  long n = p.numel();
+  auto stream = cur_stream();
+  const int nb = 256;
+  auto partial = torch::empty({nb}, p.options());
+  auto norm = torch::empty({1}, p.options());
+  hipLaunchKernelGGL(grad_norm_sq_partial_kernel, dim3(nb), dim3(256), 0, stream,
+                     g.data_ptr<float>(), n, partial.data_ptr<float>());
+  hipLaunchKernelGGL(reduce_norm_kernel, dim3(1), dim3(256), 0, stream,
+                     partial.data_ptr<float>(), nb, norm.data_ptr<float>());
+  hipLaunchKernelGGL(adamw_flat_kernel, dim3(512), dim3(256), 0, stream,
+                     p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), norm.data_ptr<float>(), t.data_ptr<int>(),
+                     (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                     (float)max_norm, n);
+  hipLaunchKernelGGL(advance_step_kernel, dim3(1), dim3(1), 0, stream,
+                     t.data_ptr<int>(), norm.data_ptr<float>());
+  return norm;
+}
+
+torch::Tensor proxqp_solve_hip(torch::Tensor H, torch::Tensor g, torch::Tensor C,
+                               torch::Tensor b, torch::Tensor l, torch::Tensor u,
+                               long iters, double rho, double sigma, double alpha) {
+  CHECK_IN(H);
+  CHECK_IN(g);
+  CHECK_IN(C);
+  CHECK_IN(b);
+  CHECK_IN(l);
+  CHECK_IN(u);
+  long M = g.size(0), nv = g.size(1), k = b.size(1);
+  auto x = torch::empty({M, nv}, g.options());
+  auto stream = cur_stream();
+  if (nv <= 32 && k <= 16) {
+    hipLaunchKernelGGL((proxqp_kernel<32, 16>), dim3(M), dim3(64), 0, stream,
+                       H.data_ptr<float>(), g.data_ptr<float>(), C.data_ptr<float>(),
+                       b.data_ptr<float>(), l.data_ptr<float>(), u.data_ptr<float>(),
+                       x.data_ptr<float>(), (int)M, (int)nv, (int)k, (int)iters,
+                       (float)rho, (float)sigma, (float)alpha);
+  } else if (nv <= 64 && k <= 32) {
+    hipLaunchKernelGGL((proxqp_kernel<64, 32>), dim3(M), dim3(64), 0, stream,
+                       H.data_ptr<float>(), g.data_ptr<float>(), C.data_ptr<float>(),
+                       b.data_ptr<float>(), l.data_ptr<float>(), u.data_ptr<float>(),
+                       x.data_ptr<float>(), (int)M, (int)nv, (int)k, (int)iters,
+                       (float)rho, (float)sigma, (float)alpha);
+  } else {
+    TORCH_CHECK(false, "proxqp_solve_hip: nv/k too large for the kernel path");
+  }
+  return x;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_adamw_step", &fused_adamw_step,
+        "flat-buffer global-norm-clip AdamW with finite guard (K13)");
+  m.def("proxqp_solve", &proxqp_solve_hip, "batched dense QP, one wave per problem (K11)");
   m.def("gemm_bias_act", &gemm_bias_act, "Y = act(X@W + b), MFMA bf16");
   m.def("act_bwd", &act_bwd, "dZ = dY * act'(Y)");
   m.def("gemm_tn", &gemm_tn, "dW = X^T dZ, db = colsum dZ (deterministic)");

@@ -43,8 +43,18 @@ def proxqp_solve(
     Applies OSQP-style Ruiz equilibration + cost scaling first — essential
     because the GCBF+ QP mixes O(1) action costs with the 1e3 relaxation
     penalty (gcbf_plus.py:299 relax_penalty).
+
+    GPU: one-wave-per-QP HIP kernel (K11) when sizes fit; torch path is the
+    CPU oracle and the large-problem fallback.
     """
     M, n = g.shape
+    if g.is_cuda and n <= 64 and b.shape[1] <= 32:
+        from . import _require_ext
+
+        ext = _require_ext()
+        f32c = lambda t: t.to(torch.float32).contiguous()
+        return ext.proxqp_solve(f32c(H), f32c(g), f32c(C), f32c(b), f32c(l), f32c(u),
+                                iters, rho, sigma, alpha)
     k = b.shape[1]
     dtype = torch.float32
     H = H.to(dtype)

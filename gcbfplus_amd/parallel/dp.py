@@ -78,3 +78,14 @@ def all_agree(flag: bool) -> bool:
         t = t.to("cuda")
     dist.all_reduce(t, op=dist.ReduceOp.MIN)
     return bool(t.item())
+
+
+@torch.no_grad()
+def allreduce_mean_flat(flats: List[torch.Tensor]):
+    """Mean all-reduce of pre-flattened grad buckets (one call per bucket)."""
+    if not is_active():
+        return
+    w = world_size()
+    for f in flats:
+        dist.all_reduce(f, op=dist.ReduceOp.SUM)
+        f /= w

@@ -17,6 +17,8 @@ setup(
                 "gcbfplus_amd/ops/hip/gemm.hip",
                 "gcbfplus_amd/ops/hip/softmax_aggr.hip",
                 "gcbfplus_amd/ops/hip/raytrace.hip",
+                "gcbfplus_amd/ops/hip/proxqp.hip",
+                "gcbfplus_amd/ops/hip/optimizer.hip",
                 "gcbfplus_amd/ops/hip/bindings.hip",
             ],
             extra_compile_args={"cxx": ["-O3"], "nvcc": ["-O3"]},

@@ -42,6 +42,10 @@ def test_gemm_bias_act_matches_fp32(M, K, N, act):
 
 
 def test_fused_linear_backward_matches_fp32():
+    """Backward GEMMs vs an fp32 oracle that uses the SAME forward rounding
+    (kernel's y for the relu mask, bf16-rounded dz) — isolates kernel math
+    from the inherent bf16 relu-boundary mask flips; plus a loose check vs
+    the pure-fp32 chain."""
     torch.manual_seed(1)
     M, K, N = 777, 128, 256
     x = _rand(M, K, scale=0.5).requires_grad_(True)
@@ -51,16 +55,28 @@ def test_fused_linear_backward_matches_fp32():
     g = torch.randn_like(y)
     y.backward(g)
 
+    # same-rounding oracle in fp32 math
+    x_bf = x.detach().to(torch.bfloat16).float()
+    w_bf = w.detach().to(torch.bfloat16).float()
+    dz = (g * (y.detach() > 0)).to(torch.bfloat16).float()
+    dx_ref = (dz @ w_bf.t()).cpu()
+    dw_ref = (x_bf.t() @ dz).cpu()
+    db_ref = dz.sum(0).cpu()
+    for got, ref in ((x.grad, dx_ref), (w.grad, dw_ref), (b.grad, db_ref)):
+        err = (got.float().cpu() - ref).abs()
+        denom = ref.abs().mean().clamp_min(1e-3)
+        assert (err.mean() / denom) < 5e-3, (err.mean(), denom, err.max())
+
+    # loose end-to-end vs pure fp32 chain (precision-policy check)
     xf = x.detach().float().cpu().requires_grad_(True)
     wf = w.detach().float().cpu().requires_grad_(True)
     bf = b.detach().float().cpu().requires_grad_(True)
     yf = torch.relu(xf @ wf + bf)
     yf.backward(g.float().cpu())
-
     for got, ref in ((x.grad, xf.grad), (w.grad, wf.grad), (b.grad, bf.grad)):
         got = got.float().cpu()
         denom = ref.abs().mean().clamp_min(1e-3)
-        assert ((got - ref).abs().mean() / denom) < 2e-2, (got - ref).abs().max()
+        assert ((got - ref).abs().mean() / denom) < 8e-2
 
 
 def test_gemm_tn_deterministic():
@@ -175,3 +191,73 @@ def test_gpu_update_step_finite():
     ro = collect_rollout(env, algo.step, g)
     info = algo.update(ro, 0)
     assert all(np.isfinite(v) for v in info.values()), info
+
+
+def test_proxqp_kernel_matches_torch_oracle():
+    """K11 HIP solver vs the torch path (same algorithm) on CBF-QP-shaped
+    problems."""
+    from gcbfplus_amd.ops.qp import proxqp_solve
+
+    rng = np.random.default_rng(7)
+    N, nu = 8, 2
+    nv = N * nu + N
+    M = 64
+    H = np.tile(np.eye(nv, dtype=np.float32), (M, 1, 1))
+    H[:, N * nu:, N * nu:] *= 10.0
+    u_ref = rng.uniform(-1, 1, size=(M, N * nu)).astype(np.float32)
+    g = np.concatenate([-u_ref, 1e3 * np.ones((M, N), np.float32)], axis=1)
+    Lg = rng.normal(size=(M, N, N * nu)).astype(np.float32)
+    C = -np.concatenate([Lg, np.tile(np.eye(N, dtype=np.float32), (M, 1, 1))], axis=2)
+    b = (rng.normal(size=(M, N)) * 0.5).astype(np.float32)
+    l = np.concatenate([-np.ones((M, N * nu), np.float32), np.zeros((M, N), np.float32)], 1)
+    u = np.concatenate([np.ones((M, N * nu), np.float32),
+                        np.full((M, N), np.inf, np.float32)], 1)
+    ts = [torch.from_numpy(t) for t in (H, g, C, b, l, u)]
+    x_cpu = proxqp_solve(*ts, iters=100)
+    x_gpu = proxqp_solve(*[t.cuda() for t in ts], iters=100).cpu()
+    # identical algorithm; compare objectives and feasibility rather than
+    # iterates (different fp orders)
+    def obj(x):
+        return 0.5 * torch.einsum("mi,mij,mj->m", x, ts[0], x) + (ts[1] * x).sum(1)
+    og, oc = obj(x_gpu), obj(x_cpu)
+    assert (og - oc).abs().max() < 1e-2, (og - oc).abs().max()
+    viol = torch.einsum("mkn,mn->mk", ts[2], x_gpu) - ts[3]
+    assert viol.max() < 1e-3
+    assert (x_gpu >= ts[4] - 1e-4).all()
+
+
+def test_fused_adamw_matches_torch():
+    from gcbfplus_amd.ops.optim import FusedAdamW
+
+    torch.manual_seed(7)
+    net1 = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.Linear(32, 4)).cuda()
+    net2 = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.Linear(32, 4)).cuda()
+    net2.load_state_dict(net1.state_dict())
+    opt1 = FusedAdamW(net1, lr=1e-2, weight_decay=1e-3, max_grad_norm=2.0)
+    opt2 = torch.optim.AdamW(net2.parameters(), lr=1e-2, weight_decay=1e-3)
+    x = torch.randn(64, 16, device="cuda")
+    for _ in range(5):
+        opt1.zero_grad()
+        net1(x).square().mean().backward()
+        opt1.step()
+
+        opt2.zero_grad()
+        loss = net2(x).square().mean()
+        loss.backward()
+        torch.nn.utils.clip_grad_norm_(net2.parameters(), 2.0)
+        opt2.step()
+    for p1, p2 in zip(net1.parameters(), net2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5), (p1 - p2).abs().max()
+
+
+def test_fused_adamw_skips_nonfinite():
+    from gcbfplus_amd.ops.optim import FusedAdamW
+
+    net = torch.nn.Linear(8, 8).cuda()
+    opt = FusedAdamW(net, lr=1e-2, max_grad_norm=2.0)
+    before = opt.pflat.clone()
+    opt.zero_grad()
+    opt.gflat.fill_(float("nan"))
+    opt.step()
+    assert torch.equal(opt.pflat, before)
+    assert opt.t.item() == 0
