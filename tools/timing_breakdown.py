@@ -1,4 +1,5 @@
 """Per-phase wall timing of one GCBF+ training step on GPU (dev tool)."""
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time
 
 import numpy as np
