@@ -35,7 +35,7 @@ class GCBFPlus(GCBF):
         self.cbf_tgt = copy.deepcopy(self.cbf)
         for p in self.cbf_tgt.parameters():
             p.requires_grad_(False)
-        self.qp_iters = 100
+        self.qp_iters = 150
         self.qp_relax_penalty = 1e3
 
     def _create_optimizers(self):

@@ -92,22 +92,42 @@ def proxqp_solve(
     los = lo * E
     his = hi * E
 
-    K = Hs + sigma * eye_n + rho * As.transpose(1, 2) @ As
-    Lc = torch.linalg.cholesky(K)
+    AtA = As.transpose(1, 2) @ As
+    rho_v = torch.full((M, 1), rho, dtype=dtype, device=g.device)
 
+    def factor(rv):
+        K = Hs + sigma * eye_n + rv[:, :, None] * AtA
+        return torch.linalg.cholesky(K)
+
+    Lc = factor(rho_v)
     x = torch.zeros(M, n, dtype=dtype, device=g.device)
     z = torch.zeros(M, m_c, dtype=dtype, device=g.device)
     y = torch.zeros(M, m_c, dtype=dtype, device=g.device)
 
-    for _ in range(iters):
-        rhs = sigma * x - gs + torch.einsum("mkn,mk->mn", As, rho * z - y)
+    for it in range(iters):
+        rhs = sigma * x - gs + torch.einsum("mkn,mk->mn", As, rho_v * z - y)
         xt = torch.cholesky_solve(rhs.unsqueeze(-1), Lc).squeeze(-1)
         zt = torch.einsum("mkn,mn->mk", As, xt)
         x = alpha * xt + (1 - alpha) * x
         z_relax = alpha * zt + (1 - alpha) * z
-        znew = torch.clamp(z_relax + y / rho, los, his)
-        y = y + rho * (z_relax - znew)
+        znew = torch.clamp(z_relax + y / rho_v, los, his)
+        y = y + rho_v * (z_relax - znew)
         z = znew
+        # OSQP-style per-problem rho adaptation every 25 iterations
+        if (it + 1) % 25 == 0 and it + 1 < iters:
+            ax = torch.einsum("mkn,mn->mk", As, x)
+            pri = (ax - z).abs().amax(dim=1)
+            pri_den = torch.maximum(ax.abs().amax(1), z.abs().amax(1)).clamp_min(1e-8)
+            hx = torch.einsum("mij,mj->mi", Hs, x)
+            aty = torch.einsum("mkn,mk->mn", As, y)
+            dua = (hx + gs + aty).abs().amax(dim=1)
+            dua_den = torch.maximum(
+                torch.maximum(hx.abs().amax(1), aty.abs().amax(1)), gs.abs().amax(1)
+            ).clamp_min(1e-8)
+            ratio = torch.sqrt((pri / pri_den).clamp_min(1e-10) /
+                               (dua / dua_den).clamp_min(1e-10))
+            rho_v = (rho_v * ratio[:, None]).clamp(1e-6, 1e6)
+            Lc = factor(rho_v)
     x = x * D
 
     # ---- polish (OSQP-style): exact solve on the active set --------------
@@ -116,32 +136,54 @@ def proxqp_solve(
     # equality-KKT solve up to O(1/mu); fall back to the ADMM iterate where
     # the polish is worse (non-PSD corner cases).
     Af = A.to(torch.float64)
-    zf = torch.einsum("mkn,mn->mk", Af, x.to(torch.float64))
-    tol = 1e-4
-    at_lo = zf <= (lo.to(torch.float64) + tol)
-    at_hi = zf >= (hi.to(torch.float64) - tol)
-    active = (at_lo | at_hi) | (y.abs() > 1e-6 * y.abs().amax(dim=1, keepdim=True))
-    vbound = torch.where(at_hi, hi.to(torch.float64), lo.to(torch.float64))
-    vbound = torch.where(active & torch.isfinite(vbound), vbound, zf)
-    w = active.to(torch.float64)
-    mu = 1e8
+    lof = lo.to(torch.float64)
+    hif = hi.to(torch.float64)
     Hf = H.to(torch.float64)
     gf = g.to(torch.float64)
-    Kp = Hf + mu * torch.einsum("mki,mk,mkj->mij", Af, w, Af)
-    rp = -gf + mu * torch.einsum("mki,mk,mk->mi", Af, w, vbound)
-    xp = torch.linalg.solve(Kp, rp)
+    mu = 1e8
+    tol = 1e-4
 
     def _score(xx):
+        """Violation-penalized merit: a tiny box/ineq violation under a
+        large linear cost (the 1e3 relax penalty) fakes a lower objective,
+        so feasibility must be priced into the comparison."""
         xx32 = xx.to(dtype)
         obj = 0.5 * torch.einsum("mi,mij,mj->m", xx32, H, xx32) + (g * xx32).sum(1)
         zz = torch.einsum("mkn,mn->mk", A, xx32)
         viol = torch.clamp(zz - hi, min=0.0) + torch.clamp(lo - zz, min=0.0)
-        return obj, viol.amax(dim=1)
+        return obj + 1e6 * viol.sum(dim=1)
 
-    obj_a, viol_a = _score(x)
-    obj_p, viol_p = _score(xp)
-    take_polish = (viol_p <= viol_a + 1e-5) & (obj_p <= obj_a + 1e-6)
-    return torch.where(take_polish[:, None], xp.to(dtype), x)
+    x_best = x
+    s_best = _score(x)
+    x_cur = x.to(torch.float64)
+    released = torch.zeros(M, m_c, dtype=torch.bool, device=g.device)
+    for pp in range(4):
+        zf = torch.einsum("mkn,mn->mk", Af, x_cur)
+        at_lo = zf <= (lof + tol)
+        at_hi = zf >= (hif - tol)  # includes rows the candidate violates
+        active = (at_lo | at_hi) & ~released
+        if pp == 0:  # first pass also trusts the ADMM duals
+            active |= y.abs() > 1e-6 * y.abs().amax(dim=1, keepdim=True)
+        vbound = torch.where(at_hi, hif, lof)
+        vbound = torch.where(active & torch.isfinite(vbound), vbound, zf)
+        w = active.to(torch.float64)
+        Kp = Hf + mu * torch.einsum("mki,mk,mkj->mij", Af, w, Af)
+        rp = -gf + mu * torch.einsum("mki,mk,mk->mi", Af, w, vbound)
+        xp = torch.linalg.solve(Kp, rp)
+        # penalty multiplier estimate lambda = mu * (A xp - vbound): wrong
+        # sign means the pin fights the KKT conditions -> release next pass
+        resid = torch.einsum("mkn,mn->mk", Af, xp) - vbound
+        released = released | (active & at_hi & ~at_lo & (resid < -1e-12)) \
+            | (active & at_lo & ~at_hi & (resid > 1e-12))
+        # project onto the box: the penalty solve leaves box-active vars
+        # biased by g/mu, which fakes a lower objective while violating box
+        xp = torch.clamp(xp, l.to(torch.float64), u.to(torch.float64).clamp(max=1e30))
+        sp = _score(xp)
+        better = sp <= s_best
+        x_best = torch.where(better[:, None], xp.to(dtype), x_best)
+        s_best = torch.minimum(sp, s_best)
+        x_cur = xp
+    return x_best
 
 
 def qp_kkt_residuals(H, g, C, b, l, u, x) -> Tuple[Tensor, Tensor]:
