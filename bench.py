@@ -56,10 +56,15 @@ def main():
         loss_h_dot_coef=0.01, max_grad_norm=2.0, seed=0,
     )
     rng = np.random.default_rng(1234 + 7919 * dp.rank())
+    graphed = None
+    if use_cuda:
+        from gcbfplus_amd.trainer.graphing import GraphedRolloutStep
+
+        graphed = GraphedRolloutStep(env, algo.step)
 
     def train_step(step):
         graph0 = env.reset(N_ENV_PER_GPU, rng)
-        rollout = collect_rollout(env, algo.step, graph0)
+        rollout = collect_rollout(env, algo.step, graph0, graphed)
         return algo.update(rollout, step)
 
     for i in range(args.warmup):

@@ -75,7 +75,15 @@ class GCBF(MultiAgentController):
         self.buffer = MaskedRolloutBuffer(size=buffer_size)
         self.unsafe_buffer = FlatSampleBuffer(size=buffer_size // 2)
         self.rng = np.random.default_rng(seed=seed + 1 + 7919 * dp.rank())
+        self._mb_graph = None
         dp.broadcast_modules([self.cbf, self.actor])
+
+    def _graphed_mb(self):
+        if self._mb_graph is None:
+            from ..trainer.graphing import GraphedMinibatchStep
+
+            self._mb_graph = GraphedMinibatchStep(self)
+        return self._mb_graph
 
     def _use_fused_optim(self) -> bool:
         return self._env.device.type == "cuda" and ops.hip_available()
@@ -221,6 +229,8 @@ class GCBF(MultiAgentController):
             chunks = torch.chunk(perm, n_mb)
             for i, mb_idx in enumerate(chunks):
                 last = ep == self.inner_epoch - 1 and i == len(chunks) - 1
+                if not last and self._graphed_mb().run(batch, mb_idx):
+                    continue
                 info = self._update_minibatch(batch[mb_idx], want_info=last)
         return info
 

@@ -98,6 +98,8 @@ class GCBFPlus(GCBF):
             chunks = torch.chunk(perm, n_mb)
             for i, mb_idx in enumerate(chunks):
                 last = ep == self.inner_epoch - 1 and i == len(chunks) - 1
+                if not last and self._graphed_mb().run(batch, mb_idx):
+                    continue
                 info = self._update_minibatch(batch[mb_idx], want_info=last)
         return info
 

@@ -105,13 +105,25 @@ class MultiAgentEnv(abc.ABC):
     @abc.abstractmethod
     def action_lim(self) -> Tuple[Tensor, Tensor]: ...
 
+    def _dev_lims(self, kind: str, device) -> Tuple[Tensor, Tensor]:
+        """Cache limit tensors per device — keeps clip ops free of
+        host->device copies (required for HIP-graph capture)."""
+        cache = getattr(self, "_lim_cache", None)
+        if cache is None:
+            cache = self._lim_cache = {}
+        key = (kind, str(device))
+        if key not in cache:
+            lo, hi = self.state_lim() if kind == "state" else self.action_lim()
+            cache[key] = (lo.to(device), hi.to(device))
+        return cache[key]
+
     def clip_state(self, state: Tensor) -> Tensor:
-        lo, hi = self.state_lim(state)
-        return torch.clamp(state, lo.to(state.device), hi.to(state.device))
+        lo, hi = self._dev_lims("state", state.device)
+        return torch.clamp(state, lo, hi)
 
     def clip_action(self, action: Tensor) -> Tensor:
-        lo, hi = self.action_lim()
-        return torch.clamp(action, lo.to(action.device), hi.to(action.device))
+        lo, hi = self._dev_lims("action", action.device)
+        return torch.clamp(action, lo, hi)
 
     # ---- core API --------------------------------------------------------
     @abc.abstractmethod

@@ -1,0 +1,144 @@
+"""HIP-graph capture of the hot loops (launch-bound on MI355X otherwise).
+
+The reference gets this from XLA tracing (one fused program per jit region);
+the MI355X build captures the eager kernel sequence once into a hipGraph and
+replays it — SURVEY.md §2.6 'one HIP graph per step'.
+
+GraphedMinibatchStep: one GCBF/GCBF+ minibatch = zero-grads + loss forward +
+backward, captured once for the fixed minibatch size; the DP all-reduce and
+the fused optimizer steps stay outside (tiny, and RCCL inside graphs is not
+worth the risk). Measured: the eager minibatch loop is ~93% of a training
+step (384 x ~14 ms of launch overhead).
+
+GraphedRolloutStep: one env step (policy forward + dynamics + LiDAR rescan +
+graph rebuild), replayed T times per rollout with ping-pong state copies.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ..utils.graph import GraphBatch
+from .data import FlatBatch
+
+
+def _capture(body) -> torch.cuda.CUDAGraph:
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):  # warmup (optimizer/grad lazily-built state)
+            body()
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        body()
+    return g
+
+
+class GraphedMinibatchStep:
+    def __init__(self, algo):
+        self.algo = algo
+        self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self.fb: Optional[FlatBatch] = None
+
+    def _alloc(self, batch: FlatBatch):
+        mb = self.algo.batch_size
+        dev = batch.states.device
+        st = torch.empty(mb, *batch.states.shape[1:], device=dev)
+        mk = torch.empty(mb, *batch.masks.shape[1:], dtype=torch.bool, device=dev)
+        sf = torch.empty(mb, *batch.safe.shape[1:], dtype=torch.bool, device=dev)
+        us = torch.empty(mb, *batch.unsafe.shape[1:], dtype=torch.bool, device=dev)
+        uq = None
+        if batch.u_qp is not None:
+            uq = torch.empty(mb, *batch.u_qp.shape[1:], device=dev)
+        self.fb = FlatBatch(st, mk, sf, us, uq)
+
+    def _body(self):
+        algo = self.algo
+        algo.cbf_optim.gflat.zero_()
+        algo.actor_optim.gflat.zero_()
+        total, _ = algo._loss(self.fb, want_info=False)
+        total.backward()
+
+    def run(self, batch: FlatBatch, idx: torch.Tensor) -> bool:
+        """Returns True if the graphed path handled this minibatch."""
+        algo = self.algo
+        if (
+            not batch.states.is_cuda
+            or idx.numel() != algo.batch_size
+            or not hasattr(algo.cbf_optim, "gflat")
+        ):
+            return False
+        if self.fb is None:
+            self._alloc(batch)
+        fb = self.fb
+        torch.index_select(batch.states, 0, idx, out=fb.states)
+        torch.index_select(batch.masks, 0, idx, out=fb.masks)
+        torch.index_select(batch.safe, 0, idx, out=fb.safe)
+        torch.index_select(batch.unsafe, 0, idx, out=fb.unsafe)
+        if fb.u_qp is not None:
+            torch.index_select(batch.u_qp, 0, idx, out=fb.u_qp)
+        if self.graph is None:
+            self.graph = _capture(self._body)
+        else:
+            self.graph.replay()
+        from ..parallel import dp
+
+        dp.allreduce_mean_flat([algo.cbf_optim.gflat, algo.actor_optim.gflat])
+        algo.cbf_optim.step()
+        algo.actor_optim.step()
+        return True
+
+
+class GraphedRolloutStep:
+    """Captures act_fn(graph) + env.step(graph, action) once; per-step replay
+    copies the previous outputs into the static inputs."""
+
+    def __init__(self, env, act_fn):
+        self.env = env
+        self.act_fn = act_fn
+        self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self.static_in: Optional[GraphBatch] = None
+        self.outs = None
+
+    def _body(self):
+        with torch.no_grad():
+            action = self.act_fn(self.static_in)
+            if isinstance(action, tuple):
+                action = action[0]
+            res = self.env.step(self.static_in, action)
+        self.outs = (action, res.graph, res.reward, res.cost, res.done)
+
+    def reset(self, graph0: GraphBatch):
+        """Prepare for a fresh batch of worlds (new obstacle tensors)."""
+        if self.static_in is None:
+            self.static_in = GraphBatch(
+                states=graph0.states.clone(),
+                mask=graph0.mask.clone(),
+                n_agents=graph0.n_agents,
+                n_rays=graph0.n_rays,
+                env_states=type(graph0.env_states)(
+                    *[f.clone() for f in graph0.env_states]
+                ),
+            )
+        else:
+            self.static_in.states.copy_(graph0.states)
+            self.static_in.mask.copy_(graph0.mask)
+            for dst, src in zip(self.static_in.env_states, graph0.env_states):
+                dst.copy_(src)
+
+    def step(self):
+        """One captured env step; returns (action, graph_out, reward, cost,
+        done) living in stable graph-pool storage (copy before next step)."""
+        if self.graph is None:
+            self.graph = _capture(self._body)
+        else:
+            self.graph.replay()
+        return self.outs
+
+    def advance(self):
+        """Copy step outputs back into the static inputs for the next step."""
+        _, g, _, _, _ = self.outs
+        self.static_in.states.copy_(g.states)
+        self.static_in.mask.copy_(g.mask)

@@ -48,6 +48,13 @@ class Trainer:
         self.update_steps = 0
         self.rng = np.random.default_rng(seed)
         self.test_rng = np.random.default_rng(seed)
+        self._ro_train = None
+        self._ro_test = None
+        if torch.cuda.is_available() and env.device.type == "cuda":
+            from .graphing import GraphedRolloutStep
+
+            self._ro_train = GraphedRolloutStep(env, algo.step)
+            self._ro_test = GraphedRolloutStep(env_test, algo.act)
 
     @staticmethod
     def _check_params(params: dict) -> dict:
@@ -75,7 +82,7 @@ class Trainer:
                     self.algo.save(self.model_dir, step)
 
             graph0 = self.env.reset(self.n_env_train, self.rng)
-            rollout = collect_rollout(self.env, self.algo.step, graph0)
+            rollout = collect_rollout(self.env, self.algo.step, graph0, self._ro_train)
             update_info = self.algo.update(rollout, step)
             self.logger.log(update_info, step=self.update_steps)
             self.update_steps += 1
@@ -85,5 +92,5 @@ class Trainer:
 
     def eval_step(self) -> dict:
         graph0 = self.env_test.reset(self.n_env_test, self.test_rng)
-        rollout = collect_rollout(self.env_test, self.algo.act, graph0)
+        rollout = collect_rollout(self.env_test, self.algo.act, graph0, self._ro_test)
         return eval_rollout_metrics(self.env_test, rollout)

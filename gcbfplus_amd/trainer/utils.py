@@ -15,11 +15,12 @@ from .data import Rollout
 
 
 @torch.no_grad()
-def collect_rollout(env, act_fn: Callable, graph0: GraphBatch) -> Rollout:
+def collect_rollout(env, act_fn: Callable, graph0: GraphBatch, graphed=None) -> Rollout:
     """Roll all B worlds forward max_episode_steps with act_fn(graph)->action.
 
     The reference scans a jitted body (trainer/utils.py:43-53); here the body
-    is a host loop of batched device kernels (captured in a HIP graph on GPU).
+    is a batch of device kernels per step, replayed from a captured HIP graph
+    when ``graphed`` (a GraphedRolloutStep) is provided on GPU.
     """
     B, T = graph0.batch_size, env.max_episode_steps
     V, S = graph0.n_nodes, graph0.state_dim
@@ -31,6 +32,23 @@ def collect_rollout(env, act_fn: Callable, graph0: GraphBatch) -> Rollout:
     rewards = torch.empty(B, T, device=dev)
     costs = torch.empty(B, T, device=dev)
     dones = torch.zeros(B, T, dtype=torch.bool, device=dev)
+
+    if graphed is not None and graph0.states.is_cuda:
+        graphed.reset(graph0)
+        for t in range(T):
+            states[:, t] = graphed.static_in.states
+            masks[:, t] = graphed.static_in.mask
+            action, _, reward, cost, done = graphed.step()
+            actions[:, t] = action
+            rewards[:, t] = reward
+            costs[:, t] = cost
+            dones[:, t] = done
+            graphed.advance()
+        return Rollout(
+            states=states, masks=masks, actions=actions, rewards=rewards, costs=costs,
+            dones=dones, next_states=graphed.static_in.states.clone(),
+            next_mask=graphed.static_in.mask.clone(), obstacles=graph0.env_states,
+        )
 
     graph = graph0
     for t in range(T):
