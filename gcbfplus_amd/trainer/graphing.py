@@ -63,9 +63,12 @@ class GraphedMinibatchStep:
 
     def run(self, batch: FlatBatch, idx: torch.Tensor) -> bool:
         """Returns True if the graphed path handled this minibatch."""
+        import os
+
         algo = self.algo
         if (
-            not batch.states.is_cuda
+            os.environ.get("GCBF_NO_HIPGRAPH")
+            or not batch.states.is_cuda
             or idx.numel() != algo.batch_size
             or not hasattr(algo.cbf_optim, "gflat")
         ):

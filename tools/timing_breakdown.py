@@ -78,6 +78,16 @@ def main():
         sync()
         t["polyak"] = time.perf_counter() - t0
 
+        # the real path (graphed minibatches) for comparison
+        t0 = time.perf_counter()
+        g0b = env.reset(16, rng)
+        rob = collect_rollout(env, algo.step, g0b)
+        sync()
+        tb = time.perf_counter()
+        algo.update(rob, 99)
+        sync()
+        t["FULL algo.update (graphed)"] = time.perf_counter() - tb
+
         total = sum(t.values())
         print(f"--- rep {rep}: total ~{total:.3f}s ---")
         for k, v in t.items():
