@@ -84,8 +84,8 @@ class GraphedMinibatchStep:
             torch.index_select(batch.u_qp, 0, idx, out=fb.u_qp)
         if self.graph is None:
             self.graph = _capture(self._body)
-        else:
-            self.graph.replay()
+        # stream capture records without executing -> always replay
+        self.graph.replay()
         from ..parallel import dp
 
         dp.allreduce_mean_flat([algo.cbf_optim.gflat, algo.actor_optim.gflat])
@@ -136,8 +136,7 @@ class GraphedRolloutStep:
         done) living in stable graph-pool storage (copy before next step)."""
         if self.graph is None:
             self.graph = _capture(self._body)
-        else:
-            self.graph.replay()
+        self.graph.replay()
         return self.outs
 
     def advance(self):
