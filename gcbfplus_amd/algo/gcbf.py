@@ -138,23 +138,30 @@ class GCBF(MultiAgentController):
     def _edge_feats(self, graph: GraphBatch, states: Optional[Tensor] = None) -> Tensor:
         return self._env.edge_feats(graph, states)
 
+    def _net_inputs(self, graph: GraphBatch, states: Optional[Tensor] = None):
+        """(edge_feats, msg_in): the fused layer-0 input when the config
+        allows (single GNN layer + state-diff edge family), else edge feats."""
+        if self.gnn_layers == 1 and self._env.fused_edge:
+            return None, self._env.edge_msg_in(graph, states)
+        return self._edge_feats(graph, states), None
+
     def act(self, graph: GraphBatch) -> Tensor:
         if self.online_pol_refine:
             return self.online_policy_refinement(graph)
         with torch.no_grad():
-            e = self._edge_feats(graph)
-            return 2 * self.actor(graph, e) + self._env.u_ref(graph)
+            e, mi = self._net_inputs(graph)
+            return 2 * self.actor(graph, e, msg_in=mi) + self._env.u_ref(graph)
 
     @torch.no_grad()
     def step(self, graph: GraphBatch) -> Tuple[Tensor, Tensor]:
-        e = self._edge_feats(graph)
-        action = self.actor(graph, e)
+        e, mi = self._net_inputs(graph)
+        action = self.actor(graph, e, msg_in=mi)
         log_pi = torch.zeros_like(action)
         return 2 * action + self._env.u_ref(graph), log_pi
 
     def get_cbf(self, graph: GraphBatch) -> Tensor:
-        e = self._edge_feats(graph)
-        return self.cbf(graph, e)
+        e, mi = self._net_inputs(graph)
+        return self.cbf(graph, e, msg_in=mi)
 
     def online_policy_refinement(self, graph: GraphBatch) -> Tensor:
         """Test-time gradient repair of the action (reference gcbf.py:161-201)."""
@@ -246,18 +253,18 @@ class GCBF(MultiAgentController):
         action fed to forward_graph is the RAW actor output."""
         env = self._env
         g = mb.graph(env)
-        e = self._edge_feats(g)
-        h = self.cbf(g, e).squeeze(-1).reshape(-1)
+        e, mi = self._net_inputs(g)
+        h = self.cbf(g, e, msg_in=mi).squeeze(-1).reshape(-1)
         safe_m = mb.safe.reshape(-1)
         unsafe_m = mb.unsafe.reshape(-1)
 
         loss_unsafe, acc_unsafe = _hinge_unsafe(h, unsafe_m, self.eps)
         loss_safe, acc_safe = _hinge_safe(h, safe_m, self.eps)
 
-        action = self.actor(g, e)
+        action = self.actor(g, e, msg_in=mi)
         next_g = env.forward_graph(g, action)
-        e2 = self._edge_feats(next_g)
-        h_next = self.cbf(next_g, e2).squeeze(-1).reshape(-1)
+        e2, mi2 = self._net_inputs(next_g)
+        h_next = self.cbf(next_g, e2, msg_in=mi2).squeeze(-1).reshape(-1)
         h_dot = (h_next - h) / env.dt
         val = torch.relu(-h_dot - self.alpha * h + self.eps)
         loss_h_dot = val.mean()

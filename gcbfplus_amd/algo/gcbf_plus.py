@@ -190,8 +190,8 @@ class GCBFPlus(GCBF):
     def _loss(self, mb: FlatBatch, want_info: bool = True) -> Tuple[Tensor, dict]:
         env = self._env
         g = mb.graph(env)
-        e = self._edge_feats(g)
-        h2 = self.cbf(g, e).squeeze(-1)  # (mb, N)
+        e, mi = self._net_inputs(g)
+        h2 = self.cbf(g, e, msg_in=mi).squeeze(-1)  # (mb, N)
         h = h2.reshape(-1)
         safe_m = mb.safe.reshape(-1)
         unsafe_m = mb.unsafe.reshape(-1)
@@ -200,15 +200,17 @@ class GCBFPlus(GCBF):
         loss_safe, acc_safe = _hinge_safe(h, safe_m, self.eps)
 
         # action = 2*actor + u_ref (the deployed policy)
-        action = 2 * self.actor(g, e) + env.u_ref(g)
+        action = 2 * self.actor(g, e, msg_in=mi) + env.u_ref(g)
         next_g = env.forward_graph(g, action)
-        e2 = self._edge_feats(next_g)
-        h_next = self.cbf(next_g, e2).squeeze(-1).reshape(-1)
+        e2, mi2 = self._net_inputs(next_g)
+        h_next = self.cbf(next_g, e2, msg_in=mi2).squeeze(-1).reshape(-1)
         h_dot = (h_next - h) / env.dt
 
         # stop-gradient branch: CBF params detached, actor path alive
         det_params = {k: v.detach() for k, v in self.cbf.named_parameters()}
-        h_next_ng = functional_call(self.cbf, det_params, (next_g, e2)).squeeze(-1).reshape(-1)
+        h_next_ng = functional_call(
+            self.cbf, det_params, (next_g, e2), {"msg_in": mi2}
+        ).squeeze(-1).reshape(-1)
         h_dot_ng = (h_next_ng - h.detach()) / env.dt
 
         labeled = safe_m | unsafe_m

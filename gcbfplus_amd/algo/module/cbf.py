@@ -24,7 +24,7 @@ class CBFNet(nn.Module):
         self.head = MLP(128, (256, 256), act="relu", act_final=False)
         self.out = Dense(256, 1, act="tanh")
 
-    def forward(self, graph: GraphBatch, edge_feats: Tensor) -> Tensor:
+    def forward(self, graph: GraphBatch, edge_feats: Tensor, msg_in=None) -> Tensor:
         """-> h: (B, N, 1) in [-1, 1]."""
-        x = self.gnn(graph, edge_feats)
+        x = self.gnn(graph, edge_feats, msg_in0=msg_in)
         return self.out(self.head(x))

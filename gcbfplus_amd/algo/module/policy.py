@@ -26,9 +26,9 @@ class DeterministicPolicyNet(nn.Module):
         self.head = MLP(128, (256, 256), act="relu", act_final=False)
         self.out = Dense(256, action_dim, act="tanh")
 
-    def forward(self, graph: GraphBatch, edge_feats: Tensor) -> Tensor:
+    def forward(self, graph: GraphBatch, edge_feats: Tensor, msg_in=None) -> Tensor:
         """-> raw policy output in [-1, 1]^nu, shape (B, N, nu)."""
-        x = self.gnn(graph, edge_feats)
+        x = self.gnn(graph, edge_feats, msg_in0=msg_in)
         return self.out(self.head(x))
 
 

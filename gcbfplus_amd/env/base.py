@@ -28,6 +28,7 @@ class StepResult(NamedTuple):
 
 class MultiAgentEnv(abc.ABC):
     PARAMS: dict = {}
+    fused_edge = False  # True when ops.edge_msg_in covers this env's edge features
 
     def __init__(
         self,

@@ -158,6 +158,17 @@ class DoubleIntegrator(MultiAgentEnv):
         mask = self.build_mask(states)
         return GraphBatch(states=states, mask=mask, n_agents=n, n_rays=r, env_states=obstacles)
 
+    fused_edge = True  # state-diff + pos-clip family: ops.edge_msg_in applies
+
+    def edge_msg_in(self, graph: GraphBatch, states: Optional[Tensor] = None) -> Tensor:
+        """Fused layer-0 GNN input (see ops.edge_msg_in)."""
+        from .. import ops
+
+        if states is None:
+            states = graph.states
+        return ops.edge_msg_in(states, self.num_agents, self.n_rays, self.pos_dim,
+                               self._params["comm_radius"])
+
     def edge_feats(self, graph: GraphBatch, states: Optional[Tensor] = None) -> Tensor:
         """Dense (B, N, D, edge_dim) differentiable edge features.
 

@@ -56,13 +56,15 @@ class GNNLayer(nn.Module):
         mask: Tensor,  # (B, N, D)
         send_idx: Tensor,  # (N, D)
         agents_only: bool,
+        msg_in: Tensor = None,  # optional fused (B, N, D, K[pad]) input
     ) -> Tensor:
         B, V, F = node_feats.shape
         N, D = mask.shape[1], mask.shape[2]
-        flat_idx = send_idx.reshape(-1)  # (N*D,)
-        sender = node_feats[:, flat_idx].reshape(B, N, D, F)
-        recv = node_feats[:, :N, None, :].expand(B, N, D, F)
-        msg_in = torch.cat([edge_feats, sender, recv], dim=-1)
+        if msg_in is None:
+            flat_idx = send_idx.reshape(-1)  # (N*D,)
+            sender = node_feats[:, flat_idx].reshape(B, N, D, F)
+            recv = node_feats[:, :N, None, :].expand(B, N, D, F)
+            msg_in = torch.cat([edge_feats, sender, recv], dim=-1)
         msg = self.msg_out(self.msg_mlp(msg_in))  # (B,N,D,msg_dim)
         gate = self.attn_out(self.attn_mlp(msg)).squeeze(-1)  # (B,N,D)
         aggr = ops.masked_softmax_aggr(gate, msg, mask)  # (B,N,msg_dim)
@@ -93,19 +95,26 @@ class GNN(nn.Module):
         self.node_dim = node_dim
         self.out_dim = out_dim
 
-    def forward(self, graph: GraphBatch, edge_feats: Tensor, node_feats: Optional[Tensor] = None
+    def forward(self, graph: GraphBatch, edge_feats: Optional[Tensor],
+                node_feats: Optional[Tensor] = None, msg_in0: Optional[Tensor] = None
                 ) -> Tensor:
         B = graph.batch_size
         N, R, V = graph.n_agents, graph.n_rays, graph.n_nodes
-        device = edge_feats.device
-        if node_feats is None:
-            node_feats = one_hot_node_feats(B, N, R, device, edge_feats.dtype)
-        send_idx = sender_index(N, R, device)
         n_layers = len(self.layers)
+        if edge_feats is None:
+            assert msg_in0 is not None and n_layers == 1, \
+                "edge_feats required unless a fused msg_in0 covers the single layer"
+            device = msg_in0.device
+        else:
+            device = edge_feats.device
+        if node_feats is None:
+            node_feats = one_hot_node_feats(B, N, R, device, torch.float32)
+        send_idx = sender_index(N, R, device)
         x = node_feats
         for i, layer in enumerate(self.layers):
             last = i == n_layers - 1
-            x = layer(x, edge_feats, graph.mask, send_idx, agents_only=last)
+            x = layer(x, edge_feats, graph.mask, send_idx, agents_only=last,
+                      msg_in=msg_in0 if i == 0 else None)
         return x  # (B, N, out_dim)
 
 

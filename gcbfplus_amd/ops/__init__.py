@@ -124,6 +124,12 @@ def fused_linear(x: Tensor, w: Tensor, b: Optional[Tensor], act: int = ACT_NONE)
     """
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
+    if x2.shape[-1] != w.shape[0]:
+        # zero-padded input (fused edge_msg_in emits K padded to 32): pad W
+        # rows to match — zeros x zeros contribute nothing, and autograd
+        # slices dW back to the master shape
+        assert x2.shape[-1] > w.shape[0]
+        w = torch.cat([w, w.new_zeros(x2.shape[-1] - w.shape[0], w.shape[1])], dim=0)
     if x2.is_cuda:
         y = _FusedLinearHIP.apply(x2, w, b, act)
     else:
@@ -177,6 +183,73 @@ def masked_softmax_aggr(gate: Tensor, msg: Tensor, mask: Tensor) -> Tensor:
     denom = e.sum(dim=-1, keepdim=True).clamp_min(1e-20)
     attn = e / denom
     return torch.einsum("bnd,bndc->bnc", attn.to(msg.dtype), msg)
+
+
+# --------------------------------------------------------------------------
+# edge_msg_in: fused first-layer GNN input (K2/K15)
+# --------------------------------------------------------------------------
+class _EdgeMsgInHIP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, states: Tensor, n_agents: int, n_rays: int, pos_dim: int,
+                k_pad: int, comm: float):
+        ext = _require_ext()
+        st = states.contiguous()
+        X = ext.edge_msg_in_fwd(st, n_agents, n_rays, pos_dim, k_pad, comm)
+        ctx.save_for_backward(st)
+        ctx.meta = (n_agents, n_rays, pos_dim, comm)
+        return X
+
+    @staticmethod
+    def backward(ctx, dX: Tensor):
+        ext = _require_ext()
+        (st,) = ctx.saved_tensors
+        n, r, pdim, comm = ctx.meta
+        dstates = ext.edge_msg_in_bwd(st, dX.contiguous().to(torch.bfloat16), n, r,
+                                      pdim, comm)
+        return dstates, None, None, None, None, None
+
+
+def edge_msg_in(states: Tensor, n_agents: int, n_rays: int, pos_dim: int,
+                comm: float, k_pad: Optional[int] = None) -> Tensor:
+    """Fused layer-0 GNN input: per edge slot
+    [clip(recv - send) | sender one-hot | recv one-hot | 0 pad], (B, N, D, KP).
+
+    Valid for the 'state-diff + position-clip' edge-feature family
+    (SI/DI/LinearDrone; reference double_integrator.py:275-286 + one-hot
+    node feats :288-295). GPU: bf16 kernel pair; CPU: composed fp32 torch.
+    """
+    B, V, S = states.shape
+    K = S + 6
+    if k_pad is None:
+        k_pad = (K + 31) // 32 * 32
+    if states.is_cuda:
+        return _EdgeMsgInHIP.apply(states, n_agents, n_rays, pos_dim, k_pad, comm)
+    # CPU compose (fp32)
+    n, r = n_agents, n_rays
+    recv = states[:, :n, None, :]
+    senders = torch.cat(
+        [
+            states[:, None, :n].expand(B, n, n, S),
+            states[:, n : 2 * n, None, :],
+            states[:, 2 * n :].reshape(B, n, r, S),
+        ],
+        dim=2,
+    )
+    e = recv - senders
+    pos = e[..., :pos_dim]
+    nrm = torch.sqrt(1e-6 + (pos * pos).sum(-1, keepdim=True))
+    coef = torch.where(nrm > comm, comm / torch.clamp(nrm, min=comm), torch.ones_like(nrm))
+    e = torch.cat([pos * coef, e[..., pos_dim:]], dim=-1)
+    D = n + 1 + r
+    oh = torch.zeros(n, D, 6, dtype=states.dtype, device=states.device)
+    oh[:, :n, 2] = 1.0  # agent senders: 001
+    oh[:, n, 1] = 1.0  # goal: 010
+    oh[:, n + 1 :, 0] = 1.0  # obs: 100
+    oh[:, :, 5] = 1.0  # receiver is always an agent: 001
+    out = torch.cat([e, oh[None].expand(B, n, D, 6)], dim=-1)
+    if k_pad > K:
+        out = torch.cat([out, out.new_zeros(B, n, D, k_pad - K)], dim=-1)
+    return out
 
 
 # --------------------------------------------------------------------------

@@ -25,6 +25,8 @@ __global__ void advance_step_kernel(int*, const float*);
 template <int MAXNV, int MAXK>
 __global__ void proxqp_kernel(const float*, const float*, const float*, const float*, const float*,
                               const float*, float*, int, int, int, int, float, float, float);
+__global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float);
+__global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float);
 
 #define CHECK_IN(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 
@@ -119,7 +121,7 @@ std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
   auto stream = cur_stream();
   hipLaunchKernelGGL(gemm_tn_partial_kernel, dim3(gk, gn, S), dim3(256), 0, stream,
                      bfp(x), bfp(dz), partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
-  hipLaunchKernelGGL(colsum_partial_kernel, dim3((N + 255) / 256, S), dim3(256), 0, stream,
+  hipLaunchKernelGGL(colsum_partial_kernel, dim3((N + 31) / 32, S), dim3(32, 8), 0, stream,
                      bfp(dz), db_partial.data_ptr<float>(), M, (int)N, (int)S);
   auto dw = torch::empty({K, N}, opts);
   auto db = torch::empty({N}, opts);
@@ -237,7 +239,38 @@ torch::Tensor proxqp_solve_hip(torch::Tensor H, torch::Tensor g, torch::Tensor C
   return x;
 }
 
+torch::Tensor edge_msg_in_fwd(torch::Tensor states, long N, long R, long pdim, long KP,
+                              double comm) {
+  CHECK_IN(states);
+  long B = states.size(0), V = states.size(1), S = states.size(2);
+  long D = N + 1 + R;
+  TORCH_CHECK(S <= 16 && KP <= 64);
+  auto X = torch::empty({B, N, D, KP}, states.options().dtype(torch::kBFloat16));
+  long total = B * N * D;
+  hipLaunchKernelGGL(edge_msg_in_fwd_kernel, dim3((total + 255) / 256), dim3(256), 0,
+                     cur_stream(), states.data_ptr<float>(), bfp_mut(X), (int)B, (int)N,
+                     (int)R, (int)S, (int)pdim, (int)KP, (float)comm);
+  return X;
+}
+
+torch::Tensor edge_msg_in_bwd(torch::Tensor states, torch::Tensor dX, long N, long R,
+                              long pdim, double comm) {
+  CHECK_IN(states);
+  CHECK_IN(dX);
+  long B = states.size(0), V = states.size(1), S = states.size(2);
+  long KP = dX.size(-1);
+  auto dstates = torch::empty_like(states);
+  long total = B * V;
+  hipLaunchKernelGGL(edge_msg_in_bwd_kernel, dim3((total + 255) / 256), dim3(256), 0,
+                     cur_stream(), states.data_ptr<float>(), bfp(dX),
+                     dstates.data_ptr<float>(), (int)B, (int)N, (int)R, (int)S, (int)pdim,
+                     (int)KP, (float)comm);
+  return dstates;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("edge_msg_in_fwd", &edge_msg_in_fwd);
+  m.def("edge_msg_in_bwd", &edge_msg_in_bwd);
   m.def("fused_adamw_step", &fused_adamw_step,
         "flat-buffer global-norm-clip AdamW with finite guard (K13)");
   m.def("proxqp_solve", &proxqp_solve_hip, "batched dense QP, one wave per problem (K11)");

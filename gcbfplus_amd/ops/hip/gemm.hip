@@ -241,29 +241,47 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
     }
 }
 
-// partials (S,K,N) f32 -> dW (K,N) f32 (fixed-order sum: deterministic)
+// partials (S,K,N) f32 -> dW (K,N) f32 (fixed-order sum: deterministic).
+// 4 independent accumulators break the serial-add latency chain.
 __global__ void reduce_partials_kernel(const float* __restrict__ partial,
                                        float* __restrict__ dW, long KN, int S) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= KN) return;
-  float a = 0.f;
-  for (int s = 0; s < S; ++s) a += partial[(long)s * KN + i];
-  dW[i] = a;
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int s = 0;
+  for (; s + 4 <= S; s += 4) {
+    a0 += partial[(long)s * KN + i];
+    a1 += partial[(long)(s + 1) * KN + i];
+    a2 += partial[(long)(s + 2) * KN + i];
+    a3 += partial[(long)(s + 3) * KN + i];
+  }
+  for (; s < S; ++s) a0 += partial[(long)s * KN + i];
+  dW[i] = (a0 + a1) + (a2 + a3);
 }
 
 // db[n] = sum_m dZ[m][n]: split-M partials (grid.y = S) then reduce via
-// reduce_partials_kernel. Deterministic (fixed split boundaries + order).
+// reduce_partials_kernel. Deterministic (fixed split boundaries, fixed
+// tree order). Block = 32 columns x 8 row-threads, LDS tree per column.
 __global__ void colsum_partial_kernel(const bf16_t* __restrict__ dZ, float* __restrict__ partial,
                                       long M, int N, int S) {
-  const int n = blockIdx.x * blockDim.x + threadIdx.x;
+  __shared__ float red[8][32];
+  const int c = threadIdx.x;       // 0..31 column lane
+  const int r = threadIdx.y;       // 0..7 row group
+  const int n = blockIdx.x * 32 + c;
   const int s = blockIdx.y;
-  if (n >= N) return;
   const long m_per = (M + S - 1) / S;
   const long ms = (long)s * m_per;
   const long me = (ms + m_per < M) ? ms + m_per : M;
   float a = 0.f;
-  for (long m = ms; m < me; ++m) a += (float)dZ[m * N + n];
-  partial[(long)s * N + n] = a;
+  if (n < N)
+    for (long m = ms + r; m < me; m += 8) a += (float)dZ[m * N + n];
+  red[r][c] = a;
+  __syncthreads();
+  if (r == 0 && n < N) {
+    float t0 = red[0][c] + red[1][c], t1 = red[2][c] + red[3][c];
+    float t2 = red[4][c] + red[5][c], t3 = red[6][c] + red[7][c];
+    partial[(long)s * N + n] = (t0 + t1) + (t2 + t3);
+  }
 }
 
 // ---------------------------------------------------------------------------

@@ -18,6 +18,7 @@ setup(
                 "gcbfplus_amd/ops/hip/softmax_aggr.hip",
                 "gcbfplus_amd/ops/hip/raytrace.hip",
                 "gcbfplus_amd/ops/hip/proxqp.hip",
+                "gcbfplus_amd/ops/hip/edge_msg.hip",
                 "gcbfplus_amd/ops/hip/optimizer.hip",
                 "gcbfplus_amd/ops/hip/bindings.hip",
             ],

@@ -324,3 +324,29 @@ def test_graphed_rollout_matches_eager():
     ro_eager2 = collect_rollout(env, algo.step, g2)
     ro_graph2 = collect_rollout(env, algo.step, g2, graphed)
     assert torch.allclose(ro_eager2.states, ro_graph2.states, atol=1e-5)
+
+
+def test_edge_msg_in_matches_cpu_compose():
+    from gcbfplus_amd.env import make_env
+
+    torch.manual_seed(13)
+    env = make_env("DoubleIntegrator", num_agents=6, area_size=3.0, max_step=4,
+                   device="cpu")
+    g = env.reset(3, np.random.default_rng(7))
+    mi_cpu = ops.edge_msg_in(g.states, 6, env.n_rays, 2, env.params["comm_radius"])
+    mi_gpu = ops.edge_msg_in(g.states.cuda(), 6, env.n_rays, 2,
+                             env.params["comm_radius"]).float().cpu()
+    err = (mi_gpu - mi_cpu).abs()
+    assert err.max() < 5e-3, err.max()  # bf16 storage of f32 values
+
+    # backward: fused vjp vs autograd through the CPU compose
+    st = g.states.cuda().requires_grad_(True)
+    mi = ops.edge_msg_in(st, 6, env.n_rays, 2, env.params["comm_radius"])
+    gout = torch.randn_like(mi.float())
+    mi.backward(gout.to(torch.bfloat16))
+    st_c = g.states.clone().requires_grad_(True)
+    mi_c = ops.edge_msg_in(st_c, 6, env.n_rays, 2, env.params["comm_radius"])
+    mi_c.backward(gout.to(torch.bfloat16).float().cpu())
+    err = (st.grad.cpu() - st_c.grad).abs()
+    denom = st_c.grad.abs().mean().clamp_min(1e-4)
+    assert (err.mean() / denom) < 2e-2, (err.mean(), err.max(), denom)
