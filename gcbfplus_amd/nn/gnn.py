@@ -23,9 +23,18 @@ from ..utils.graph import GraphBatch
 from .mlp import MLP, Dense
 
 
+_SENDER_IDX_CACHE = {}
+_ONEHOT_CACHE = {}
+
+
 def sender_index(n_agents: int, n_rays: int, device) -> Tensor:
     """(N, D) node indices of the sender in each edge slot (computable gather
-    pattern — slot d of receiver i): agents | own goal | own lidar hits."""
+    pattern — slot d of receiver i): agents | own goal | own lidar hits.
+    Cached per shape/device (graph-capture friendly, zero per-call kernels)."""
+    key = (n_agents, n_rays, str(device))
+    hit = _SENDER_IDX_CACHE.get(key)
+    if hit is not None:
+        return hit
     n, r = n_agents, n_rays
     d = n + 1 + r
     idx = torch.empty(n, d, dtype=torch.long, device=device)
@@ -33,6 +42,7 @@ def sender_index(n_agents: int, n_rays: int, device) -> Tensor:
     idx[:, n] = torch.arange(n, device=device) + n
     base = 2 * n + torch.arange(n, device=device)[:, None] * r
     idx[:, n + 1 :] = base + torch.arange(r, device=device)[None, :]
+    _SENDER_IDX_CACHE[key] = idx
     return idx
 
 
@@ -120,10 +130,14 @@ class GNN(nn.Module):
 
 def one_hot_node_feats(B: int, N: int, R: int, device, dtype=torch.float32) -> Tensor:
     """Constant node features: agent=001, goal=010, obstacle=100
-    (reference env/double_integrator.py:288-295)."""
-    V = 2 * N + N * R
-    f = torch.zeros(V, 3, device=device, dtype=dtype)
-    f[:N, 2] = 1.0
-    f[N : 2 * N, 1] = 1.0
-    f[2 * N :, 0] = 1.0
-    return f[None].expand(B, V, 3)
+    (reference env/double_integrator.py:288-295). Cached per shape/device."""
+    key = (N, R, str(device), dtype)
+    f = _ONEHOT_CACHE.get(key)
+    if f is None:
+        V = 2 * N + N * R
+        f = torch.zeros(V, 3, device=device, dtype=dtype)
+        f[:N, 2] = 1.0
+        f[N : 2 * N, 1] = 1.0
+        f[2 * N :, 0] = 1.0
+        _ONEHOT_CACHE[key] = f
+    return f[None].expand(B, f.shape[0], 3)

@@ -8,10 +8,13 @@
 typedef __bf16 bf16_t_;
 template <int ACT>
 __global__ void gemm_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
+template <int ACT>
+__global__ void gemm_bias_act_sm_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
+__global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int);
 template <int ACT, typename OutT>
 __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, OutT*, int, int, int);
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
-__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, float*, int, int, int, int);
+__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 __global__ void reduce_partials_kernel(const float*, float*, long, int);
 __global__ void colsum_partial_kernel(const bf16_t_*, float*, long, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
@@ -80,7 +83,18 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
     return y;
   }
   auto y = torch::empty({M, N}, x.options());
-  {
+  if (M <= 16384) {
+    // small-M tile: 32x64 so mid-size layers still fill 256 CUs
+    dim3 grid((M + 31) / 32, (N + 63) / 64);
+    size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 32 * 40 * sizeof(uint16_t);
+    auto launch = [&](auto kernel) {
+      hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(x), bfp(w),
+                         bias.data_ptr<float>(), bfp_mut(y), (int)M, (int)N, (int)K);
+    };
+    if (act == 0) launch(gemm_bias_act_sm_kernel<0>);
+    else if (act == 1) launch(gemm_bias_act_sm_kernel<1>);
+    else launch(gemm_bias_act_sm_kernel<2>);
+  } else {
     dim3 grid((M + 127) / 128, (N + 63) / 64);
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 128 * 40 * sizeof(uint16_t);
     auto launch = [&](auto kernel) {
@@ -120,15 +134,13 @@ std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
   auto db_partial = torch::empty({S, N}, opts);
   auto stream = cur_stream();
   hipLaunchKernelGGL(gemm_tn_partial_kernel, dim3(gk, gn, S), dim3(256), 0, stream,
-                     bfp(x), bfp(dz), partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
-  hipLaunchKernelGGL(colsum_partial_kernel, dim3((N + 31) / 32, S), dim3(32, 8), 0, stream,
-                     bfp(dz), db_partial.data_ptr<float>(), M, (int)N, (int)S);
+                     bfp(x), bfp(dz), partial.data_ptr<float>(), db_partial.data_ptr<float>(),
+                     (int)M, (int)N, (int)K, (int)S);
   auto dw = torch::empty({K, N}, opts);
   auto db = torch::empty({N}, opts);
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3((K * N + 255) / 256), dim3(256), 0, stream,
-                     partial.data_ptr<float>(), dw.data_ptr<float>(), K * N, (int)S);
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3((N + 255) / 256), dim3(256), 0, stream,
-                     db_partial.data_ptr<float>(), db.data_ptr<float>(), N, (int)S);
+  hipLaunchKernelGGL(reduce_dw_db_kernel, dim3((K * N + N + 255) / 256), dim3(256), 0, stream,
+                     partial.data_ptr<float>(), db_partial.data_ptr<float>(),
+                     dw.data_ptr<float>(), db.data_ptr<float>(), K * N, (int)N, (int)S);
   return {dw, db};
 }
 

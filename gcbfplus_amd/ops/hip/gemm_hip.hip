@@ -110,6 +110,88 @@ void gemm_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict
 }
 
 // ---------------------------------------------------------------------------
+// Small-M variant: BM=32 x BN=64 tile (4 waves as 2x2, wave 16x32) so
+// mid-size rows (update/head layers, M ~ 2k) still fill all 256 CUs.
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void gemm_bias_act_sm_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
+                             const float* __restrict__ bias, bf16_t* __restrict__ Y,
+                             int M, int N, int K) {
+  constexpr int BM = 32, BN = 64, BK = 32, APAD = 40;
+  extern __shared__ char smem[];
+  bf16_t* sB = (bf16_t*)smem;                                        // [K/8][BN][8]
+  bf16_t* sA = (bf16_t*)(smem + (K / 8) * BN * 8 * sizeof(bf16_t));  // [BM][APAD]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wm = w >> 1, wn = w & 1;  // wave tile: 16(M) x 32(N)
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  for (int c = tid; c < K * 8; c += 256) {
+    const int k = c >> 3;
+    const int co = (c & 7) * 8;
+    bf16_t v[8];
+    if (n0 + co + 7 < N) {
+      *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
+  }
+
+  f32x4 acc[1][2];
+  acc[0][0] = f32x4{0.f, 0.f, 0.f, 0.f};
+  acc[0][1] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int arow = tid >> 3;           // 32 rows x 8 threads per row
+  const int acol = (tid & 7) * 4;      // 4 bf16 (8 B) per thread
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    bf16_t av[4];
+    const long arow_g = (long)(m0 + arow);
+    if (arow_g < M) {
+      *(uint2*)av = *(const uint2*)(X + arow_g * K + k0 + acol);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) av[i] = (bf16_t)0.f;
+    }
+    *(uint2*)(sA + arow * APAD + acol) = *(uint2*)av;
+    __syncthreads();
+
+    bf16x8 afr = *(const bf16x8*)(sA + (wm * 16 + (lane & 15)) * APAD + (lane >> 4) * 8);
+    bf16x8 bfr[2];
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      bfr[nf] = *(const bf16x8*)(sB + (((k0 >> 3) + (lane >> 4)) * BN + wn * 32 + nf * 16 + (lane & 15)) * 8);
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      acc[0][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr, bfr[nf], acc[0][nf], 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    const int col = n0 + wn * 32 + nf * 16 + (lane & 15);
+    if (col >= N) continue;
+    const float bv = bias[col];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long row = m0 + wm * 16 + (lane >> 4) * 4 + r;
+      if (row < M) Y[row * N + col] = (bf16_t)apply_act(acc[0][nf][r] + bv, ACT);
+    }
+  }
+}
+
+template __global__ void gemm_bias_act_sm_kernel<0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+
+// ---------------------------------------------------------------------------
 // Small-N path (N <= 16): one wave per output row, W cached in LDS.
 // grid: ceil(M/4); block 256 (4 waves).
 // ---------------------------------------------------------------------------
@@ -156,10 +238,12 @@ __global__ void act_bwd_kernel(const bf16_t* __restrict__ dY, const bf16_t* __re
 // ---------------------------------------------------------------------------
 __launch_bounds__(256) __global__
 void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
-                            float* __restrict__ partial, int M, int N, int K, int S) {
+                            float* __restrict__ partial, float* __restrict__ db_partial,
+                            int M, int N, int K, int S) {
   constexpr int BKDIM = 64, BN = 64, BMR = 32, TPAD = 40;
   __shared__ bf16_t sXT[BKDIM][TPAD];   // [k][m] transposed X tile
   __shared__ bf16_t sB[BMR / 8][BN][8]; // dZ tile, m-blocked
+  __shared__ float sDb[4][BN];          // db tree reduce (k0==0 blocks)
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int w = tid >> 6;
@@ -178,6 +262,8 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
   for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float db_acc = 0.f;  // column sum of dZ (blockIdx.x == 0 only)
+  const int db_c = tid & 63, db_q = tid >> 6;
 
   for (long m0 = ms; m0 < me; m0 += BMR) {
     __syncthreads();
@@ -213,6 +299,12 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
     }
     __syncthreads();
 
+    if (blockIdx.x == 0) {
+      // free db: the dZ tile is already staged; rows q*8..q*8+7 of col c
+      const bf16x8 v = *(const bf16x8*)(&sB[db_q][db_c][0]);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
+    }
     bf16x8 afr[2], bfr[2];
 #pragma unroll
     for (int kf = 0; kf < 2; ++kf)
@@ -227,6 +319,13 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
         acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
   }
 
+  if (blockIdx.x == 0) {
+    sDb[db_q][db_c] = db_acc;
+    __syncthreads();
+    if (db_q == 0 && n0 + db_c < N)
+      db_partial[(long)s * N + n0 + db_c] =
+          (sDb[0][db_c] + sDb[1][db_c]) + (sDb[2][db_c] + sDb[3][db_c]);
+  }
   float* out = partial + (long)s * K * N;
 #pragma unroll
   for (int kf = 0; kf < 2; ++kf)
@@ -240,6 +339,31 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
         if (krow < K) out[(long)krow * N + col] = acc[kf][nf][r];
       }
     }
+}
+
+// (dw_partial (S,K,N), db_partial (S,N)) -> (dW, db) in ONE launch
+// (fixed-order sums: deterministic; 4 accumulators hide add latency).
+__global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* __restrict__ pb,
+                                    float* __restrict__ dW, float* __restrict__ db,
+                                    long KN, int N, int S) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const bool is_db = i >= KN;
+  if (i >= KN + N) return;
+  const float* src = is_db ? pb : pw;
+  const long stride = is_db ? N : KN;
+  const long off = is_db ? i - KN : i;
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int s = 0;
+  for (; s + 4 <= S; s += 4) {
+    a0 += src[(long)s * stride + off];
+    a1 += src[(long)(s + 1) * stride + off];
+    a2 += src[(long)(s + 2) * stride + off];
+    a3 += src[(long)(s + 3) * stride + off];
+  }
+  for (; s < S; ++s) a0 += src[(long)s * stride + off];
+  const float r = (a0 + a1) + (a2 + a3);
+  if (is_db) db[off] = r;
+  else dW[off] = r;
 }
 
 // partials (S,K,N) f32 -> dW (K,N) f32 (fixed-order sum: deterministic).
