@@ -169,7 +169,7 @@ class GCBFPlus(GCBF):
         M, N, S = graph.batch_size, self.n_agents, graph.state_dim
         states = graph.states.detach()
         with torch.enable_grad():
-            if self.gnn_layers == 1:
+            if self.gnn_layers == 1 and getattr(env, "analytic_edge_jac", True):
                 e = env.edge_feats(graph, states).detach().requires_grad_(True)
                 h = cbf_net(graph, e).squeeze(-1)  # (M, N)
                 (ge,) = torch.autograd.grad(h.sum(), e)  # (M, N, D, E)

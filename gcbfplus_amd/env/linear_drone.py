@@ -107,7 +107,7 @@ class LinearDrone(DoubleIntegrator):
             rng, batch, self.num_agents, 3, self.area_size, inside_np,
             min_dist=4 * self._params["drone_radius"], max_travel=self.max_travel,
         )
-        zeros = np.zeros_like(starts)
+        zeros = np.zeros((batch, self.num_agents, self.state_dim - 3), dtype=np.float32)
         agent = torch.from_numpy(np.concatenate([starts, zeros], -1)).to(self.device)
         goal = torch.from_numpy(np.concatenate([goals, zeros], -1)).to(self.device)
         obstacles = Sphere(*[t.to(self.device) for t in obs_cpu])

@@ -267,3 +267,41 @@ def test_dubins_update_smoke():
     ro = collect_rollout(env, algo.step, g)
     info = algo.update(ro, 0)
     assert np.isfinite(info["loss/total"])
+
+
+def test_crazyflie_env_basics():
+    from gcbfplus_amd.env import make_env
+
+    env = make_env("CrazyFlie", num_agents=2, area_size=2.0, max_step=8, device="cpu")
+    rng = np.random.default_rng(0)
+    g = env.reset(2, rng)
+    assert g.states.shape[2] == 12
+    # hover equilibrium: zero action from rest holds position exactly
+    st = torch.zeros(1, 2, 12)
+    st[:, :, :3] = 1.0
+    nxt = env.agent_step_rk4(st, torch.zeros(1, 2, 4))
+    assert (nxt[..., :3] - st[..., :3]).abs().max() < 1e-5
+    # u_ref flies toward the goal
+    gg = g[0:1]
+    d0 = torch.linalg.vector_norm(gg.agent_states[..., :3] - gg.goal_states[..., :3], dim=-1).mean()
+    for _ in range(100):
+        gg = env.step(gg, env.u_ref(gg)).graph
+    d1 = torch.linalg.vector_norm(gg.agent_states[..., :3] - gg.goal_states[..., :3], dim=-1).mean()
+    assert d1 < d0 * 0.5
+
+
+def test_crazyflie_update_smoke():
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.algo import make_algo
+    from gcbfplus_amd.trainer.utils import collect_rollout
+
+    torch.manual_seed(5)
+    env = make_env("CrazyFlie", num_agents=2, area_size=2.0, max_step=4, device="cpu")
+    algo = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim, n_agents=2,
+                     gnn_layers=1, batch_size=4, buffer_size=8, horizon=2,
+                     inner_epoch=1, seed=0)
+    g = env.reset(2, np.random.default_rng(4))
+    ro = collect_rollout(env, algo.step, g)
+    info = algo.update(ro, 0)
+    assert np.isfinite(info["loss/total"])
