@@ -68,20 +68,36 @@ def test(args):
     n_epi = args.epi
     for epi in range(n_epi):
         graph0 = env.reset(1, rng)
-        rollout = collect_rollout(env, act_fn, graph0)
-        g = rollout.graph_at(env)
-        T = rollout.time_horizon
-        coll = env.collision_mask(g).reshape(1, T, -1)
-        finish = env.finish_mask(g).reshape(1, T, -1)
-        a_safe = 1.0 - coll.amax(dim=1).float()  # (1, N)
-        a_finish = finish.amax(dim=1).float()
+        if args.nojit_rollout:
+            # streamed path for large swarms (reference env/base.py:191-259):
+            # step-by-step metrics, no stored graphs/edges
+            g = graph0
+            ever_coll = env.collision_mask(g).float()
+            ever_fin = env.finish_mask(g).float()
+            with torch.no_grad():
+                for _ in range(env.max_episode_steps):
+                    a = act_fn(g)
+                    g = env.step(g, a).graph
+                    ever_coll = torch.maximum(ever_coll, env.collision_mask(g).float())
+                    ever_fin = torch.maximum(ever_fin, env.finish_mask(g).float())
+            a_safe, a_finish = 1.0 - ever_coll, ever_fin
+            rollout = None
+        else:
+            rollout = collect_rollout(env, act_fn, graph0)
+            g = rollout.graph_at(env)
+            T = rollout.time_horizon
+            coll = env.collision_mask(g).reshape(1, T, -1)
+            finish = env.finish_mask(g).reshape(1, T, -1)
+            a_safe = 1.0 - coll.amax(dim=1).float()  # (1, N)
+            a_finish = finish.amax(dim=1).float()
         a_success = a_safe * a_finish
         safe_rates.append(a_safe.mean().item())
         finish_rates.append(a_finish.mean().item())
         success_rates.append(a_success.mean().item())
         print(f"epi {epi}: safe {a_safe.mean():.3f} finish {a_finish.mean():.3f} "
               f"success {a_success.mean():.3f}")
-        rollouts.append(rollout)
+        if rollout is not None:
+            rollouts.append(rollout)
 
     print(
         f"safe rate: {100*np.mean(safe_rates):.3f}%, "
@@ -131,6 +147,8 @@ def main():
     parser.add_argument("--cpu", action="store_true", default=False)
     parser.add_argument("--log", type=str, default=None)
     parser.add_argument("--cbf", type=int, default=None)
+    parser.add_argument("--nojit-rollout", action="store_true", default=False,
+                        help="streamed eval without storing rollouts (512+ agents)")
     parser.add_argument("--no-video", action="store_true", default=False)
     parser.add_argument("--max-videos", type=int, default=1)
     parser.add_argument("--debug", action="store_true", default=False)

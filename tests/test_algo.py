@@ -129,3 +129,24 @@ def test_online_policy_refinement_runs(small_setup):
         algo.online_pol_refine = False
     assert a.shape == (1, env.num_agents, env.action_dim)
     assert torch.isfinite(a).all()
+
+
+def test_value_net_and_tanh_normal():
+    from gcbfplus_amd.algo.module.value import ValueNet
+    from gcbfplus_amd.algo.module.policy import TanhNormalPolicyNet
+    from gcbfplus_amd.env import make_env
+
+    torch.manual_seed(0)
+    env = make_env("DoubleIntegrator", num_agents=3, area_size=2.0, max_step=4,
+                   device="cpu")
+    g = env.reset(2, np.random.default_rng(0))
+    e = env.edge_feats(g)
+    v = ValueNet(env.node_dim, env.edge_dim)(g, e)
+    assert v.shape == (2,) and torch.isfinite(v).all()
+    pol = TanhNormalPolicyNet(env.node_dim, env.edge_dim, env.action_dim)
+    a, logp = pol.sample(g, e)
+    assert a.shape == (2, 3, 2) and (a.abs() <= 1).all()
+    lp = pol.log_prob(g, e, a)
+    assert torch.isfinite(lp).all()
+    m = pol.mode(g, e)
+    assert m.shape == a.shape
