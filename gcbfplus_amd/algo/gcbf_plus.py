@@ -19,6 +19,7 @@ import torch
 from torch import Tensor
 from torch.func import functional_call
 
+from .. import ops
 from ..ops.qp import proxqp_solve
 from ..parallel import dp
 from ..trainer.data import FlatBatch, Rollout
@@ -210,10 +211,6 @@ class GCBFPlus(GCBF):
         h_both = self.cbf(big, e_big, msg_in=mi_big).squeeze(-1)  # (2B, N)
         h = h_both[:B].reshape(-1)
         h_next = h_both[B:].reshape(-1)
-        h_dot = (h_next - h) / env.dt
-
-        loss_unsafe, acc_unsafe = _hinge_unsafe(h, unsafe_m, self.eps)
-        loss_safe, acc_safe = _hinge_safe(h, safe_m, self.eps)
 
         # stop-gradient branch: CBF params detached, actor path alive; the
         # VALUE equals h_next (same params) but gradients route differently
@@ -223,7 +220,31 @@ class GCBFPlus(GCBF):
         h_next_ng = functional_call(
             self.cbf, det_params, (next_g, e2), {"msg_in": mi2}
         ).squeeze(-1).reshape(-1)
+
+        if g.states.is_cuda and ops.hip_available():
+            # fused loss kernel (K10): identical math, 2 kernels vs ~160
+            nu = action.shape[-1]
+            total, parts = ops.gcbf_plus_loss(
+                h, h_next, h_next_ng, action.reshape(-1, nu), mb.u_qp.reshape(-1, nu),
+                safe_m, unsafe_m, env.dt, self.alpha, self.eps,
+                self.loss_action_coef, self.loss_unsafe_coef, self.loss_safe_coef,
+                self.loss_h_dot_coef,
+            )
+            info = {}
+            if want_info:
+                p = parts.tolist()
+                info = {
+                    "loss/action": p[0], "loss/unsafe": p[1], "loss/safe": p[2],
+                    "loss/h_dot": p[3], "loss/total": float(total.detach()),
+                    "acc/unsafe": p[4], "acc/safe": p[5], "acc/h_dot": p[6],
+                    "acc/unsafe_data_ratio": p[7],
+                }
+            return total, info
+
+        h_dot = (h_next - h) / env.dt
         h_dot_ng = (h_next_ng - h.detach()) / env.dt
+        loss_unsafe, acc_unsafe = _hinge_unsafe(h, unsafe_m, self.eps)
+        loss_safe, acc_safe = _hinge_safe(h, safe_m, self.eps)
 
         labeled = safe_m | unsafe_m
         val = torch.relu(-h_dot - self.alpha * h + self.eps)

@@ -263,3 +263,42 @@ def raytrace_rect(pos: Tensor, points: Tensor, n_rays: int, sense_range: float) 
         _require_ext()
         return _EXT.raytrace_rect(pos.contiguous(), points.contiguous(), n_rays, sense_range)
     raise NotImplementedError("CPU path goes through env.get_lidar composition")
+
+
+# --------------------------------------------------------------------------
+# fused GCBF+ loss (K10)
+# --------------------------------------------------------------------------
+class _GCBFLossHIP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, h, h_next, h_ng, action, u_qp, safe, unsafe, params):
+        ext = _require_ext()
+        dt, alpha, eps, c_act, c_unsafe, c_safe, c_hdot = params
+        args = [t.contiguous() for t in (h, h_next, h_ng, action, u_qp)]
+        out = ext.gcbf_loss_fwd(*args, safe.contiguous(), unsafe.contiguous(),
+                                dt, alpha, eps, c_act, c_unsafe, c_safe, c_hdot)
+        ctx.save_for_backward(*args, safe, unsafe, out)
+        ctx.params = params
+        parts = out[1:9]
+        ctx.mark_non_differentiable(parts)
+        return out[0], parts
+
+    @staticmethod
+    def backward(ctx, d_total, _d_parts):
+        ext = _require_ext()
+        h, h_next, h_ng, action, u_qp, safe, unsafe, out = ctx.saved_tensors
+        dt, alpha, eps, c_act, c_unsafe, c_safe, c_hdot = ctx.params
+        dh, dh_next, dh_ng, daction = ext.gcbf_loss_bwd(
+            h, h_next, h_ng, action, u_qp, safe, unsafe, out,
+            d_total.reshape(1).contiguous().float(),
+            dt, alpha, eps, c_act, c_unsafe, c_safe, c_hdot,
+        )
+        return dh, dh_next, dh_ng, daction, None, None, None, None
+
+
+def gcbf_plus_loss(h, h_next, h_ng, action, u_qp, safe, unsafe, dt, alpha, eps,
+                   c_act, c_unsafe, c_safe, c_hdot):
+    """Fused GCBF+ loss (reference gcbf_plus.py:364-431). Returns
+    (total scalar, parts (8,) detached: [action, unsafe, safe, h_dot losses,
+    acc_unsafe, acc_safe, acc_h_dot, unsafe_ratio])."""
+    return _GCBFLossHIP.apply(h, h_next, h_ng, action, u_qp, safe, unsafe,
+                              (dt, alpha, eps, c_act, c_unsafe, c_safe, c_hdot))

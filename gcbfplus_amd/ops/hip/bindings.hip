@@ -30,6 +30,8 @@ __global__ void proxqp_kernel(const float*, const float*, const float*, const fl
                               const float*, float*, int, int, int, int, float, float, float);
 __global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float);
 __global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float);
+__global__ void gcbf_loss_fwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, float*, long, int, float, float, float, float, float, float, float);
+__global__ void gcbf_loss_bwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, const float*, const float*, float*, float*, float*, float*, long, int, float, float, float, float, float, float, float);
 
 #define CHECK_IN(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 
@@ -124,7 +126,7 @@ std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
   CHECK_IN(dz);
   long M = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == M);
-  long gk = (K + 63) / 64, gn = (N + 63) / 64;
+  long gk = (K + 127) / 128, gn = (N + 127) / 128;
   // deterministic split count: aim for ~512 blocks, depends on shapes only
   long S = std::min<long>(64, std::max<long>(1, 512 / std::max<long>(1, gk * gn)));
   S = std::min<long>(S, std::max<long>(1, (M + 31) / 32));
@@ -280,7 +282,54 @@ torch::Tensor edge_msg_in_bwd(torch::Tensor states, torch::Tensor dX, long N, lo
   return dstates;
 }
 
+torch::Tensor gcbf_loss_fwd(torch::Tensor h, torch::Tensor h_next, torch::Tensor h_ng,
+                            torch::Tensor action, torch::Tensor u_qp, torch::Tensor safe,
+                            torch::Tensor unsafe, double dt, double alpha, double eps,
+                            double c_act, double c_unsafe, double c_safe, double c_hdot) {
+  CHECK_IN(h);
+  CHECK_IN(h_next);
+  CHECK_IN(h_ng);
+  CHECK_IN(action);
+  CHECK_IN(u_qp);
+  long n = h.numel();
+  int nu = action.numel() / n;
+  auto out = torch::empty({11}, h.options());
+  hipLaunchKernelGGL(gcbf_loss_fwd_kernel, dim3(1), dim3(256), 0, cur_stream(),
+                     h.data_ptr<float>(), h_next.data_ptr<float>(), h_ng.data_ptr<float>(),
+                     action.data_ptr<float>(), u_qp.data_ptr<float>(),
+                     safe.data_ptr<bool>(), unsafe.data_ptr<bool>(), out.data_ptr<float>(),
+                     n, nu, (float)dt, (float)alpha, (float)eps, (float)c_act,
+                     (float)c_unsafe, (float)c_safe, (float)c_hdot);
+  return out;
+}
+
+std::vector<torch::Tensor> gcbf_loss_bwd(torch::Tensor h, torch::Tensor h_next,
+                                         torch::Tensor h_ng, torch::Tensor action,
+                                         torch::Tensor u_qp, torch::Tensor safe,
+                                         torch::Tensor unsafe, torch::Tensor out,
+                                         torch::Tensor gscale, double dt, double alpha,
+                                         double eps, double c_act, double c_unsafe,
+                                         double c_safe, double c_hdot) {
+  long n = h.numel();
+  int nu = action.numel() / n;
+  auto dh = torch::empty_like(h);
+  auto dh_next = torch::empty_like(h);
+  auto dh_ng = torch::empty_like(h);
+  auto daction = torch::empty_like(action);
+  hipLaunchKernelGGL(gcbf_loss_bwd_kernel, dim3(32), dim3(256), 0, cur_stream(),
+                     h.data_ptr<float>(), h_next.data_ptr<float>(), h_ng.data_ptr<float>(),
+                     action.data_ptr<float>(), u_qp.data_ptr<float>(),
+                     safe.data_ptr<bool>(), unsafe.data_ptr<bool>(), out.data_ptr<float>(),
+                     gscale.data_ptr<float>(), dh.data_ptr<float>(),
+                     dh_next.data_ptr<float>(), dh_ng.data_ptr<float>(),
+                     daction.data_ptr<float>(), n, nu, (float)dt, (float)alpha, (float)eps,
+                     (float)c_act, (float)c_unsafe, (float)c_safe, (float)c_hdot);
+  return {dh, dh_next, dh_ng, daction};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gcbf_loss_fwd", &gcbf_loss_fwd);
+  m.def("gcbf_loss_bwd", &gcbf_loss_bwd);
   m.def("edge_msg_in_fwd", &edge_msg_in_fwd);
   m.def("edge_msg_in_bwd", &edge_msg_in_bwd);
   m.def("fused_adamw_step", &fused_adamw_step,

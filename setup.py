@@ -19,6 +19,7 @@ setup(
                 "gcbfplus_amd/ops/hip/raytrace.hip",
                 "gcbfplus_amd/ops/hip/proxqp.hip",
                 "gcbfplus_amd/ops/hip/edge_msg.hip",
+                "gcbfplus_amd/ops/hip/loss.hip",
                 "gcbfplus_amd/ops/hip/optimizer.hip",
                 "gcbfplus_amd/ops/hip/bindings.hip",
             ],

@@ -350,3 +350,51 @@ def test_edge_msg_in_matches_cpu_compose():
     err = (st.grad.cpu() - st_c.grad).abs()
     denom = st_c.grad.abs().mean().clamp_min(1e-4)
     assert (err.mean() / denom) < 2e-2, (err.mean(), err.max(), denom)
+
+
+def test_fused_loss_matches_eager_math():
+    """K10 fused loss vs a CPU fp32 transcription of gcbf_plus.py:364-431,
+    values AND gradients."""
+    torch.manual_seed(21)
+    B, N, nu = 64, 8, 2
+    n = B * N
+    dt, alpha, eps = 0.03, 1.0, 0.02
+    coefs = (1e-4, 1.0, 1.0, 0.01)
+    h = (torch.randn(n, device="cuda") * 0.1).requires_grad_(True)
+    h_next = (h.detach() + torch.randn(n, device="cuda") * 0.01).requires_grad_(True)
+    h_ng = h_next.detach().clone().requires_grad_(True)
+    action = torch.randn(n, nu, device="cuda").requires_grad_(True)
+    u_qp = torch.randn(n, nu, device="cuda")
+    safe = torch.rand(n, device="cuda") < 0.5
+    unsafe = ~safe & (torch.rand(n, device="cuda") < 0.3)
+
+    total, parts = ops.gcbf_plus_loss(h, h_next, h_ng, action, u_qp, safe, unsafe,
+                                      dt, alpha, eps, *coefs)
+    total.backward()
+
+    def ref(hc, hnc, hgc, ac):
+        h_dot = (hnc - hc) / dt
+        h_dot_ng = (hgc - hc.detach()) / dt
+        h_unsafe = torch.where(unsafe.cpu(), hc, torch.full_like(hc, -2 * eps))
+        loss_unsafe = torch.relu(h_unsafe + eps).sum() / (unsafe.float().sum().cpu() + 1e-6)
+        h_safe = torch.where(safe.cpu(), hc, torch.full_like(hc, 2 * eps))
+        loss_safe = torch.relu(-h_safe + eps).sum() / (safe.float().sum().cpu() + 1e-6)
+        labeled = (safe | unsafe).cpu()
+        val = torch.relu(-h_dot - alpha * hc + eps)
+        val_ng = torch.relu(-h_dot_ng - alpha * hc + eps)
+        loss_h_dot = torch.where(labeled, val, val_ng).mean()
+        loss_action = (ac - u_qp.cpu()).square().sum(-1).mean()
+        return (coefs[0] * loss_action + coefs[1] * loss_unsafe
+                + coefs[2] * loss_safe + coefs[3] * loss_h_dot)
+
+    hc = h.detach().cpu().requires_grad_(True)
+    hnc = h_next.detach().cpu().requires_grad_(True)
+    hgc = h_ng.detach().cpu().requires_grad_(True)
+    ac = action.detach().cpu().requires_grad_(True)
+    t_ref = ref(hc, hnc, hgc, ac)
+    t_ref.backward()
+
+    assert abs(total.item() - t_ref.item()) < 1e-4, (total.item(), t_ref.item())
+    for got, want in ((h.grad, hc.grad), (h_next.grad, hnc.grad),
+                      (h_ng.grad, hgc.grad), (action.grad, ac.grad)):
+        assert torch.allclose(got.cpu(), want, atol=1e-6), (got.cpu() - want).abs().max()
