@@ -126,7 +126,7 @@ std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
   CHECK_IN(dz);
   long M = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == M);
-  long gk = (K + 127) / 128, gn = (N + 127) / 128;
+  long gk = (K + 63) / 64, gn = (N + 63) / 64;
   // deterministic split count: aim for ~512 blocks, depends on shapes only
   long S = std::min<long>(64, std::max<long>(1, 512 / std::max<long>(1, gk * gn)));
   S = std::min<long>(S, std::max<long>(1, (M + 31) / 32));

@@ -239,38 +239,36 @@ __launch_bounds__(256) __global__
 void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
                             float* __restrict__ partial, float* __restrict__ db_partial,
                             int M, int N, int K, int S) {
-  // 128x128 output tile (4 waves as 2x2, wave 64x64 = 4x4 MFMA frags):
-  // halves the X and dZ re-read traffic vs 64x64 tiles.
-  constexpr int BKDIM = 128, BN = 128, BMR = 32, TPAD = 40;
+  constexpr int BKDIM = 64, BN = 64, BMR = 32, TPAD = 40;
   __shared__ bf16_t sXT[BKDIM][TPAD];   // [k][m] transposed X tile
   __shared__ bf16_t sB[BMR / 8][BN][8]; // dZ tile, m-blocked
   __shared__ float sDb[4][BN];          // db tree reduce (k0==0 blocks)
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int w = tid >> 6;
-  const int wk = w >> 1, wn = w & 1;  // wave tile 64(K) x 64(N)
+  const int wk = w >> 1, wn = w & 1;  // wave tile 32(K) x 32(N)
   const int k0 = blockIdx.x * BKDIM;
   const int n0 = blockIdx.y * BN;
   const int s = blockIdx.z;
 
+  // split-M range for this s (fixed boundaries -> deterministic)
   const long m_per = ((long)M + S - 1) / S;
   const long ms = (long)s * m_per;
   const long me = (ms + m_per < (long)M) ? ms + m_per : (long)M;
 
-  f32x4 acc[4][4];
+  f32x4 acc[2][2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
-  float db_acc[2] = {0.f, 0.f};  // 2 columns per thread (BN=128, 256 thr -> 2 rows of 8)
-  const int db_c = tid & 127, db_q = tid >> 7;
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float db_acc = 0.f;  // column sum of dZ (blockIdx.x == 0 only)
+  const int db_c = tid & 63, db_q = tid >> 6;
 
   for (long m0 = ms; m0 < me; m0 += BMR) {
     __syncthreads();
-    // stage X^T: chunk c -> row m0 + (c>>4), k-chunk (c&15)*8; 32x16 chunks
-    for (int c = tid; c < BMR * 16; c += 256) {
-      const long mr = m0 + (c >> 4);
-      const int kc = (c & 15) * 8;
+    {  // stage X^T: thread t loads X[m0 + t>>3][k0 + (t&7)*8 ..+8]
+      const long mr = m0 + (tid >> 3);
+      const int kc = (tid & 7) * 8;
       bf16_t v[8];
       if (mr < me && k0 + kc + 7 < K) {
         *(bf16x8*)v = *(const bf16x8*)(X + mr * K + k0 + kc);
@@ -280,12 +278,12 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
           v[i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
       }
 #pragma unroll
-      for (int i = 0; i < 8; ++i) sXT[kc + i][c >> 4] = v[i];
+      for (int i = 0; i < 8; ++i) sXT[kc + i][tid >> 3] = v[i];
     }
-    {  // stage dZ m-blocked: chunk c -> mrow = c>>4, col8 = (c&15)*8
-      for (int c = tid; c < BMR * 16; c += 256) {
-        const long mr = m0 + (c >> 4);
-        const int co = (c & 15) * 8;
+    {  // stage dZ m-blocked: chunk c -> mrow = c>>3, col8 = (c&7)*8
+      for (int c = tid; c < BMR * 8; c += 256) {
+        const long mr = m0 + (c >> 3);
+        const int co = (c & 7) * 8;
         bf16_t v[8];
         if (mr < me && n0 + co + 7 < N) {
           *(bf16x8*)v = *(const bf16x8*)(dZ + mr * N + n0 + co);
@@ -295,38 +293,33 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
             v[i] = (mr < me && n0 + co + i < N) ? dZ[mr * N + n0 + co + i] : (bf16_t)0.f;
         }
 #pragma unroll
-        for (int i = 0; i < 8; ++i) sB[(c >> 4) >> 3][co + i][(c >> 4) & 7] = v[i];
+        for (int i = 0; i < 8; ++i) sB[(c >> 3) >> 3][co + i][(c >> 3) & 7] = v[i];
       }
     }
     __syncthreads();
 
     if (blockIdx.x == 0) {
-      // free db: dZ tile already staged; rows (db_q*2+h)*8..+8 of col db_c
+      // free db: the dZ tile is already staged; rows q*8..q*8+7 of col c
+      const bf16x8 v = *(const bf16x8*)(&sB[db_q][db_c][0]);
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        const bf16x8 v = *(const bf16x8*)(&sB[db_q * 2 + h][db_c][0]);
-#pragma unroll
-        for (int i = 0; i < 8; ++i) db_acc[h] += (float)v[i];
-      }
+      for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
     }
-    bf16x8 afr[4], bfr[4];
+    bf16x8 afr[2], bfr[2];
 #pragma unroll
-    for (int kf = 0; kf < 4; ++kf)
-      afr[kf] = *(const bf16x8*)(&sXT[wk * 64 + kf * 16 + (lane & 15)][(lane >> 4) * 8]);
+    for (int kf = 0; kf < 2; ++kf)
+      afr[kf] = *(const bf16x8*)(&sXT[wk * 32 + kf * 16 + (lane & 15)][(lane >> 4) * 8]);
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf)
-      bfr[nf] = *(const bf16x8*)(&sB[lane >> 4][wn * 64 + nf * 16 + (lane & 15)][0]);
+    for (int nf = 0; nf < 2; ++nf)
+      bfr[nf] = *(const bf16x8*)(&sB[lane >> 4][wn * 32 + nf * 16 + (lane & 15)][0]);
 #pragma unroll
-    for (int kf = 0; kf < 4; ++kf)
+    for (int kf = 0; kf < 2; ++kf)
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
+      for (int nf = 0; nf < 2; ++nf)
         acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
   }
 
   if (blockIdx.x == 0) {
-    // db_acc[h] belongs to column db_c, row-group db_q*2+h
-    sDb[db_q * 2 + 0][db_c] = db_acc[0];
-    sDb[db_q * 2 + 1][db_c] = db_acc[1];
+    sDb[db_q][db_c] = db_acc;
     __syncthreads();
     if (db_q == 0 && n0 + db_c < N)
       db_partial[(long)s * N + n0 + db_c] =
@@ -334,14 +327,14 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
   }
   float* out = partial + (long)s * K * N;
 #pragma unroll
-  for (int kf = 0; kf < 4; ++kf)
+  for (int kf = 0; kf < 2; ++kf)
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
-      const int col = n0 + wn * 64 + nf * 16 + (lane & 15);
+    for (int nf = 0; nf < 2; ++nf) {
+      const int col = n0 + wn * 32 + nf * 16 + (lane & 15);
       if (col >= N) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int krow = k0 + wk * 64 + kf * 16 + (lane >> 4) * 4 + r;
+        const int krow = k0 + wk * 32 + kf * 16 + (lane >> 4) * 4 + r;
         if (krow < K) out[(long)krow * N + col] = acc[kf][nf][r];
       }
     }
