@@ -85,7 +85,11 @@ def pwise_cbf(env, graph: GraphBatch, k: int = 3) -> Tuple[Tensor, Tensor, Tenso
     from ..env.linear_drone import LinearDrone
     from ..env.single_integrator import SingleIntegrator
 
+    from ..env.crazyflie import CrazyFlie
+
     name = type(env).__name__
+    if isinstance(env, CrazyFlie):
+        return _pwise_cbf_crazyflie(env, graph, k)
     if isinstance(env, SingleIntegrator):
         ag, sel, idx, isobs = _neighbors(graph, 2, k)
         r = env.params["car_radius"]
@@ -136,3 +140,48 @@ def pwise_cbf(env, graph: GraphBatch, k: int = 3) -> Tuple[Tensor, Tensor, Tenso
             isobs,
         )
     raise NotImplementedError(f"pwise CBF not implemented for {name}")
+
+
+def _pwise_cbf_crazyflie(env, graph: GraphBatch, k: int):
+    """3rd-order chain h2 = h1' + 50 h1, h1 = h0' + 30 h0 with the drift-only
+    12-state dynamics (reference algo/utils.py:182-300). Derivatives via
+    torch.func (the reference nests jax.jacfwd); the top-level jacobian for
+    the QP comes from one more jacrev. Eval-only path — not on the training
+    hot loop."""
+    import torch.func as tf
+
+    B, N = graph.batch_size, graph.n_agents
+    S = graph.state_dim
+    ag, sel, idx, isobs = _neighbors(graph, 3, k)
+    r = env.params["drone_radius"]
+    f = env._f_batched  # pure torch drift
+
+    def h0(x, ox):
+        return ((x[:3] - ox[:, :3]) ** 2).sum(-1) - 4 * r * r
+
+    def h1(x, ox):
+        jx = tf.jacrev(h0, argnums=0)(x, ox)  # (k, 12)
+        jo = tf.jacrev(h0, argnums=1)(x, ox)  # (k, k, 12)
+        h0dot = jx @ f(x) + torch.einsum("abc,bc->a", jo, f(ox))
+        return h0dot + 30.0 * h0(x, ox)
+
+    def h2(x, ox):
+        jx = tf.jacrev(h1, argnums=0)(x, ox)
+        jo = tf.jacrev(h1, argnums=1)(x, ox)
+        return jx @ f(x) + torch.einsum("abc,bc->a", jo, f(ox)) + 50.0 * h1(x, ox)
+
+    flat_x = ag.reshape(B * N, S)
+    flat_ox = sel.reshape(B * N, k, S)
+    h = tf.vmap(h2)(flat_x, flat_ox).reshape(B, N, k)
+    jx, jo = tf.vmap(tf.jacrev(h2, argnums=(0, 1)))(flat_x, flat_ox)
+    jx = jx.reshape(B, N, k, S)
+    kk = torch.arange(k)
+    jo_diag = jo[:, kk, kk, :].reshape(B, N, k, S)  # h2_k depends on ox_k only
+
+    jac = torch.zeros(B, N, k, N, S, device=ag.device)
+    iidx = torch.arange(N, device=ag.device)
+    jac[:, iidx, :, iidx, :] = jx.permute(1, 0, 2, 3)
+    nb = jo_diag * (~isobs).float()[..., None]
+    jac.scatter_add_(3, idx.clamp(max=N - 1)[..., None, None].expand(B, N, k, 1, S),
+                     nb[..., None, :])
+    return h, jac, isobs
