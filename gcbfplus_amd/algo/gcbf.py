@@ -134,6 +134,32 @@ class GCBF(MultiAgentController):
         with open(os.path.join(path, "cbf.pkl"), "rb") as f:
             net_from_flax_tree(self.cbf, pickle.load(f), "CBFHead", "Dense_0")
 
+    # ---- full training resume (NOT in the reference, which saves params
+    # only — gcbf.py:344-357 / SURVEY §5.4) --------------------------------
+    def save_full(self, path: str, step: int):
+        state = {
+            "step": step,
+            "cbf": self.cbf.state_dict(),
+            "actor": self.actor.state_dict(),
+            "cbf_optim": self.cbf_optim.state_dict(),
+            "actor_optim": self.actor_optim.state_dict(),
+            "rng": self.rng.bit_generator.state,
+        }
+        if hasattr(self, "cbf_tgt"):
+            state["cbf_tgt"] = self.cbf_tgt.state_dict()
+        torch.save(state, path)
+
+    def load_full(self, path: str) -> int:
+        state = torch.load(path, map_location=self._env.device, weights_only=False)
+        self.cbf.load_state_dict(state["cbf"])
+        self.actor.load_state_dict(state["actor"])
+        self.cbf_optim.load_state_dict(state["cbf_optim"])
+        self.actor_optim.load_state_dict(state["actor_optim"])
+        self.rng.bit_generator.state = state["rng"]
+        if "cbf_tgt" in state and hasattr(self, "cbf_tgt"):
+            self.cbf_tgt.load_state_dict(state["cbf_tgt"])
+        return state["step"]
+
     # ---- acting ----------------------------------------------------------
     def _edge_feats(self, graph: GraphBatch, states: Optional[Tensor] = None) -> Tensor:
         return self._env.edge_feats(graph, states)

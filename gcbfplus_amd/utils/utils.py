@@ -1,0 +1,84 @@
+"""Small tensor/tree helpers (reference ``gcbfplus/utils/utils.py:22-171``).
+
+Most of the reference's helpers exist to make pytrees jit/vmap-friendly; in
+torch the same operations are one-liners, kept here under the same names so
+reference users find them.
+"""
+from __future__ import annotations
+
+from typing import List, Sequence
+
+import numpy as np
+import torch
+from torch import Tensor
+
+
+def merge01(x: Tensor) -> Tensor:
+    """(a, b, ...) -> (a*b, ...) (reference :22-23)."""
+    return x.reshape(x.shape[0] * x.shape[1], *x.shape[2:])
+
+
+def mask2index(mask: Tensor, n_true: int) -> Tensor:
+    """Indices of the n_true True entries (reference :74-76 via top_k)."""
+    return torch.nonzero(mask, as_tuple=False).flatten()[:n_true]
+
+
+def tree_map(fn, tree):
+    if tree is None:
+        return None
+    if isinstance(tree, Tensor):
+        return fn(tree)
+    if isinstance(tree, tuple) and hasattr(tree, "_fields"):
+        return type(tree)(*[tree_map(fn, v) for v in tree])
+    if isinstance(tree, (list, tuple)):
+        return type(tree)(tree_map(fn, v) for v in tree)
+    if isinstance(tree, dict):
+        return {k: tree_map(fn, v) for k, v in tree.items()}
+    return tree
+
+
+def tree_index(tree, idx):
+    return tree_map(lambda t: t[idx], tree)
+
+
+def tree_stack(trees: Sequence):
+    """Stack a list of congruent trees along a new leading dim (:164-171)."""
+    first = trees[0]
+    if isinstance(first, Tensor):
+        return torch.stack(list(trees), dim=0)
+    if isinstance(first, tuple) and hasattr(first, "_fields"):
+        return type(first)(*[tree_stack([t[i] for t in trees]) for i in range(len(first))])
+    raise TypeError(type(first))
+
+
+def tree_merge(trees: Sequence):
+    """Concatenate trees along dim 0 (:153-161)."""
+    first = trees[0]
+    if isinstance(first, Tensor):
+        return torch.cat(list(trees), dim=0)
+    if isinstance(first, tuple) and hasattr(first, "_fields"):
+        return type(first)(*[tree_merge([t[i] for t in trees]) for i in range(len(first))])
+    raise TypeError(type(first))
+
+
+def tree_concat_at_front(t1, t2):
+    return tree_merge([t1, t2])
+
+
+def torch2np(x):
+    """jax2np analogue (:66-71): detach to host numpy."""
+    return tree_map(lambda t: t.detach().cpu().numpy(), x)
+
+
+def np2torch(x, device="cpu"):
+    return tree_map(lambda t: t, x) if isinstance(x, Tensor) else tree_map(
+        lambda t: t, x
+    ) if x is None else _np2torch(x, device)
+
+
+def _np2torch(x, device):
+    if isinstance(x, np.ndarray):
+        return torch.from_numpy(x).to(device)
+    if isinstance(x, tuple) and hasattr(x, "_fields"):
+        return type(x)(*[_np2torch(v, device) for v in x])
+    return x
