@@ -237,6 +237,8 @@ class DoubleIntegrator(MultiAgentEnv):
 
     # ---- step / forward (reference :145-181, 340-354) ----------------------
     def step(self, graph: GraphBatch, action: Tensor) -> StepResult:
+        if graph.states.is_cuda and type(self) is DoubleIntegrator:
+            return self._step_fused(graph, action)
         agent = graph.agent_states
         goal = graph.goal_states
         obstacles = graph.env_states
@@ -248,6 +250,23 @@ class DoubleIntegrator(MultiAgentEnv):
         done = torch.zeros(agent.shape[0], dtype=torch.bool, device=agent.device)
         next_graph = self.get_graph(next_agent, goal, obstacles)
         return StepResult(next_graph, reward, cost, done, {})
+
+    def _step_fused(self, graph: GraphBatch, action: Tensor) -> StepResult:
+        """One-kernel env step on GPU (K5-K8, ops/hip/env_step.hip)."""
+        from .. import ops
+
+        ext = ops._require_ext()
+        p = self._params
+        nxt, mask, reward, cost = ext.di_env_step(
+            graph.states.contiguous(), action.contiguous(),
+            graph.env_states.points.contiguous(),
+            self._K.to(graph.device).contiguous(), self.num_agents, self.n_rays,
+            self._dt, 1.0 / p["m"], p["comm_radius"], p["car_radius"], 0.5,
+        )
+        done = torch.zeros(graph.batch_size, dtype=torch.bool, device=graph.device)
+        g = GraphBatch(states=nxt, mask=mask, n_agents=self.num_agents,
+                       n_rays=self.n_rays, env_states=graph.env_states)
+        return StepResult(g, reward, cost, done, {})
 
     def forward_graph(self, graph: GraphBatch, action: Tensor) -> GraphBatch:
         action = self.clip_action(action)

@@ -31,6 +31,9 @@ __global__ void proxqp_kernel(const float*, const float*, const float*, const fl
 __global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float);
 __global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float);
 __global__ void gcbf_loss_fwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, float*, long, int, float, float, float, float, float, float, float);
+__global__ void di_env_step_kernel(const float*, const float*, const float*, const float*,
+                                   float*, bool*, float*, float*, int, int, int, float,
+                                   float, float, float, float);
 __global__ void gcbf_loss_bwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, const float*, const float*, float*, float*, float*, float*, long, int, float, float, float, float, float, float, float);
 
 #define CHECK_IN(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
@@ -327,7 +330,33 @@ std::vector<torch::Tensor> gcbf_loss_bwd(torch::Tensor h, torch::Tensor h_next,
   return {dh, dh_next, dh_ng, daction};
 }
 
+std::vector<torch::Tensor> di_env_step(torch::Tensor states, torch::Tensor action,
+                                       torch::Tensor points, torch::Tensor Kmat, long N,
+                                       long R, double dt, double inv_m, double comm,
+                                       double car_r, double vmax) {
+  CHECK_IN(states);
+  CHECK_IN(action);
+  CHECK_IN(points);
+  CHECK_IN(Kmat);
+  long B = states.size(0), V = states.size(1), K = points.size(1);
+  long D = N + 1 + R;
+  auto nxt = torch::empty_like(states);
+  auto mask = torch::empty({B, N, D}, states.options().dtype(torch::kBool));
+  auto reward = torch::empty({B}, states.options());
+  auto cost = torch::empty({B}, states.options());
+  size_t smem = (size_t)(N * 4 + N * 2 + K * 8) * sizeof(float);
+  hipLaunchKernelGGL(di_env_step_kernel, dim3(B), dim3(256), smem, cur_stream(),
+                     states.data_ptr<float>(), action.data_ptr<float>(),
+                     points.data_ptr<float>(), Kmat.data_ptr<float>(),
+                     nxt.data_ptr<float>(), mask.data_ptr<bool>(),
+                     reward.data_ptr<float>(), cost.data_ptr<float>(), (int)N, (int)K,
+                     (int)R, (float)dt, (float)inv_m, (float)comm, (float)car_r,
+                     (float)vmax);
+  return {nxt, mask, reward, cost};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("di_env_step", &di_env_step, "fused DoubleIntegrator env step (K5-K8)");
   m.def("gcbf_loss_fwd", &gcbf_loss_fwd);
   m.def("gcbf_loss_bwd", &gcbf_loss_bwd);
   m.def("edge_msg_in_fwd", &edge_msg_in_fwd);

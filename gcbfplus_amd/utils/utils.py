@@ -71,14 +71,9 @@ def torch2np(x):
 
 
 def np2torch(x, device="cpu"):
-    return tree_map(lambda t: t, x) if isinstance(x, Tensor) else tree_map(
-        lambda t: t, x
-    ) if x is None else _np2torch(x, device)
-
-
-def _np2torch(x, device):
+    """np2jax analogue (:66-71): numpy (trees) -> device tensors."""
     if isinstance(x, np.ndarray):
         return torch.from_numpy(x).to(device)
     if isinstance(x, tuple) and hasattr(x, "_fields"):
-        return type(x)(*[_np2torch(v, device) for v in x])
+        return type(x)(*[np2torch(v, device) for v in x])
     return x
