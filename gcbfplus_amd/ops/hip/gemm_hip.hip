@@ -109,6 +109,101 @@ void gemm_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// BN=128 variant: halves the A re-read traffic for N=256 layers.
+// grid: (ceil(M/128), ceil(N/128)); 4 waves 2x2, wave tile 64x64.
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void gemm_bias_act_bn128_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
+                                const float* __restrict__ bias, bf16_t* __restrict__ Y,
+                                int M, int N, int K) {
+  constexpr int BM = 128, BN = 128, BK = 32, APAD = 40;
+  extern __shared__ char smem[];
+  bf16_t* sB = (bf16_t*)smem;                                        // [K/8][BN][8]
+  bf16_t* sA = (bf16_t*)(smem + (K / 8) * BN * 8 * sizeof(bf16_t));  // [BM][APAD]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wm = w >> 1, wn = w & 1;  // wave tile: 64(M) x 64(N)
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  for (int c = tid; c < K * 16; c += 256) {
+    const int k = c >> 4;
+    const int co = (c & 15) * 8;
+    bf16_t v[8];
+    if (n0 + co + 7 < N) {
+      *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
+  }
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int arow = tid >> 1;
+  const int acol = (tid & 1) * 16;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    bf16_t av[16];
+    const long arow_g = (long)(m0 + arow);
+    if (arow_g < M) {
+      *(bf16x8*)av = *(const bf16x8*)(X + arow_g * K + k0 + acol);
+      *(bf16x8*)(av + 8) = *(const bf16x8*)(X + arow_g * K + k0 + acol + 8);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 16; ++i) av[i] = (bf16_t)0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < 16; i += 8) *(bf16x8*)(sA + arow * APAD + acol + i) = *(bf16x8*)(av + i);
+    __syncthreads();
+
+    bf16x8 afr[4];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+      afr[mf] = *(const bf16x8*)(sA + (wm * 64 + mf * 16 + (lane & 15)) * APAD + (lane >> 4) * 8);
+    bf16x8 bfr[4];
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+      bfr[nf] = *(const bf16x8*)(sB + (((k0 >> 3) + (lane >> 4)) * BN + wn * 64 + nf * 16 + (lane & 15)) * 8);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int col = n0 + wn * 64 + nf * 16 + (lane & 15);
+      if (col >= N) continue;
+      const float bv = bias[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = m0 + wm * 64 + mf * 16 + (lane >> 4) * 4 + r;
+        if (row < M) Y[row * N + col] = (bf16_t)apply_act(acc[mf][nf][r] + bv, ACT);
+      }
+    }
+  }
+}
+
+template __global__ void gemm_bias_act_bn128_kernel<0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_bn128_kernel<1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_bn128_kernel<2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+
 // ---------------------------------------------------------------------------
 // Small-M variant: BM=32 x BN=64 tile (4 waves as 2x2, wave 16x32) so
 // mid-size rows (update/head layers, M ~ 2k) still fill all 256 CUs.
