@@ -54,7 +54,7 @@ void edge_msg_in_fwd_kernel(const float* __restrict__ states, bf16_t* __restrict
     out[S + 5] = (bf16_t)1.f;
     for (int s = S + 6; s < KP; ++s) out[s] = (bf16_t)0.f;
     bf16_t* dst = X + row * KP;
-    for (int s = 0; s < KP; ++s) dst[s] = out[s];
+    for (int s = 0; s < KP; s += 8) *(bf16x8*)(dst + s) = *(bf16x8*)(out + s);
   }
 }
 
@@ -63,7 +63,14 @@ __device__ __forceinline__ void slot_vjp(const float* recv, const float* send,
                                          const bf16_t* dx, float* acc, float sign,
                                          int S, int pdim, float comm) {
   float p2 = 1e-6f;
-  float e[16];
+  float e[16], g[16];
+  // vector-load the cotangent row (dx rows are 8-element aligned: KP % 32 == 0)
+  for (int s = 0; s < S; s += 8) {
+    const bf16x8 v = *(const bf16x8*)(dx + s);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      if (s + i < 16) g[s + i] = (float)v[i];
+  }
   for (int s = 0; s < S; ++s) {
     e[s] = recv[s] - send[s];
     if (s < pdim) p2 += e[s] * e[s];
@@ -71,14 +78,14 @@ __device__ __forceinline__ void slot_vjp(const float* recv, const float* send,
   const float n = sqrtf(p2);
   if (n > comm) {
     float gdotp = 0.f;
-    for (int s = 0; s < pdim; ++s) gdotp += (float)dx[s] * e[s];
+    for (int s = 0; s < pdim; ++s) gdotp += g[s] * e[s];
     const float inv_n = 1.f / n;
     for (int s = 0; s < pdim; ++s)
-      acc[s] += sign * comm * ((float)dx[s] * inv_n - e[s] * gdotp * inv_n * inv_n * inv_n);
+      acc[s] += sign * comm * (g[s] * inv_n - e[s] * gdotp * inv_n * inv_n * inv_n);
   } else {
-    for (int s = 0; s < pdim; ++s) acc[s] += sign * (float)dx[s];
+    for (int s = 0; s < pdim; ++s) acc[s] += sign * g[s];
   }
-  for (int s = pdim; s < S; ++s) acc[s] += sign * (float)dx[s];
+  for (int s = pdim; s < S; ++s) acc[s] += sign * g[s];
 }
 
 __launch_bounds__(256) __global__
