@@ -31,7 +31,15 @@ class DubinsCar(DoubleIntegrator):
         "n_obs": 8,
     }
 
-    fused_edge = False  # edge feats use the v-vector transform, not raw diff
+    fused_edge = True  # via the mode-1 transform in ops.edge_msg_in
+
+    def edge_msg_in(self, graph, states=None):
+        from .. import ops
+
+        if states is None:
+            states = graph.states
+        return ops.edge_msg_in(states, self.num_agents, self.n_rays, 2,
+                               self._params["comm_radius"], mode=1)
 
     def __init__(self, num_agents, area_size, max_step=256, max_travel=None, dt=0.03,
                  params=None, device=None):

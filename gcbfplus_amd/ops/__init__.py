@@ -191,26 +191,26 @@ def masked_softmax_aggr(gate: Tensor, msg: Tensor, mask: Tensor) -> Tensor:
 class _EdgeMsgInHIP(torch.autograd.Function):
     @staticmethod
     def forward(ctx, states: Tensor, n_agents: int, n_rays: int, pos_dim: int,
-                k_pad: int, comm: float):
+                k_pad: int, comm: float, mode: int):
         ext = _require_ext()
         st = states.contiguous()
-        X = ext.edge_msg_in_fwd(st, n_agents, n_rays, pos_dim, k_pad, comm)
+        X = ext.edge_msg_in_fwd(st, n_agents, n_rays, pos_dim, k_pad, comm, mode)
         ctx.save_for_backward(st)
-        ctx.meta = (n_agents, n_rays, pos_dim, comm)
+        ctx.meta = (n_agents, n_rays, pos_dim, comm, mode)
         return X
 
     @staticmethod
     def backward(ctx, dX: Tensor):
         ext = _require_ext()
         (st,) = ctx.saved_tensors
-        n, r, pdim, comm = ctx.meta
+        n, r, pdim, comm, mode = ctx.meta
         dstates = ext.edge_msg_in_bwd(st, dX.contiguous().to(torch.bfloat16), n, r,
-                                      pdim, comm)
-        return dstates, None, None, None, None, None
+                                      pdim, comm, mode)
+        return dstates, None, None, None, None, None, None
 
 
 def edge_msg_in(states: Tensor, n_agents: int, n_rays: int, pos_dim: int,
-                comm: float, k_pad: Optional[int] = None) -> Tensor:
+                comm: float, k_pad: Optional[int] = None, mode: int = 0) -> Tensor:
     """Fused layer-0 GNN input: per edge slot
     [clip(recv - send) | sender one-hot | recv one-hot | 0 pad], (B, N, D, KP).
 
@@ -223,8 +223,13 @@ def edge_msg_in(states: Tensor, n_agents: int, n_rays: int, pos_dim: int,
     if k_pad is None:
         k_pad = (K + 31) // 32 * 32
     if states.is_cuda:
-        return _EdgeMsgInHIP.apply(states, n_agents, n_rays, pos_dim, k_pad, comm)
+        return _EdgeMsgInHIP.apply(states, n_agents, n_rays, pos_dim, k_pad, comm, mode)
     # CPU compose (fp32)
+    if mode == 1:  # DubinsCar edge-state transform [x, y, v cos, v sin]
+        th, v = states[..., 2], states[..., 3]
+        states = torch.stack(
+            [states[..., 0], states[..., 1], v * torch.cos(th), v * torch.sin(th)], dim=-1
+        )
     n, r = n_agents, n_rays
     recv = states[:, :n, None, :]
     senders = torch.cat(

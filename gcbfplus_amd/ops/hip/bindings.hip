@@ -30,8 +30,8 @@ __global__ void advance_step_kernel(int*, const float*);
 template <int MAXNV, int MAXK>
 __global__ void proxqp_kernel(const float*, const float*, const float*, const float*, const float*,
                               const float*, float*, int, int, int, int, float, float, float);
-__global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float);
-__global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float);
+__global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float, int);
+__global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float, int);
 __global__ void gcbf_loss_fwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, float*, long, int, float, float, float, float, float, float, float);
 __global__ void di_env_step_kernel(const float*, const float*, const float*, const float*,
                                    float*, bool*, float*, float*, int, int, int, float,
@@ -270,7 +270,7 @@ torch::Tensor proxqp_solve_hip(torch::Tensor H, torch::Tensor g, torch::Tensor C
 }
 
 torch::Tensor edge_msg_in_fwd(torch::Tensor states, long N, long R, long pdim, long KP,
-                              double comm) {
+                              double comm, long mode) {
   CHECK_IN(states);
   long B = states.size(0), V = states.size(1), S = states.size(2);
   long D = N + 1 + R;
@@ -279,12 +279,12 @@ torch::Tensor edge_msg_in_fwd(torch::Tensor states, long N, long R, long pdim, l
   long total = B * N * D;
   hipLaunchKernelGGL(edge_msg_in_fwd_kernel, dim3((total + 255) / 256), dim3(256), 0,
                      cur_stream(), states.data_ptr<float>(), bfp_mut(X), (int)B, (int)N,
-                     (int)R, (int)S, (int)pdim, (int)KP, (float)comm);
+                     (int)R, (int)S, (int)pdim, (int)KP, (float)comm, (int)mode);
   return X;
 }
 
 torch::Tensor edge_msg_in_bwd(torch::Tensor states, torch::Tensor dX, long N, long R,
-                              long pdim, double comm) {
+                              long pdim, double comm, long mode) {
   CHECK_IN(states);
   CHECK_IN(dX);
   long B = states.size(0), V = states.size(1), S = states.size(2);
@@ -294,7 +294,7 @@ torch::Tensor edge_msg_in_bwd(torch::Tensor states, torch::Tensor dX, long N, lo
   hipLaunchKernelGGL(edge_msg_in_bwd_kernel, dim3((total + 255) / 256), dim3(256), 0,
                      cur_stream(), states.data_ptr<float>(), bfp(dX),
                      dstates.data_ptr<float>(), (int)B, (int)N, (int)R, (int)S, (int)pdim,
-                     (int)KP, (float)comm);
+                     (int)KP, (float)comm, (int)mode);
   return dstates;
 }
 

@@ -21,9 +21,23 @@ __device__ __forceinline__ int sender_node(int i, int d, int N, int R) {
   return 2 * N + i * R + (d - N - 1);
 }
 
+// edge-state transform: mode 0 = identity; mode 1 = DubinsCar
+// [x, y, v cos(theta), v sin(theta)] (reference dubins_car.py:263-275)
+__device__ __forceinline__ void edge_state_of(const float* st, float* es, int S, int mode) {
+  if (mode == 1) {
+    es[0] = st[0];
+    es[1] = st[1];
+    es[2] = st[3] * __cosf(st[2]);
+    es[3] = st[3] * __sinf(st[2]);
+  } else {
+    for (int s = 0; s < S; ++s) es[s] = st[s];
+  }
+}
+
 __launch_bounds__(256) __global__
 void edge_msg_in_fwd_kernel(const float* __restrict__ states, bf16_t* __restrict__ X,
-                            int B, int N, int R, int S, int pdim, int KP, float comm) {
+                            int B, int N, int R, int S, int pdim, int KP, float comm,
+                            int mode) {
   const int D = N + 1 + R;
   const int V = 2 * N + N * R;
   const long total = (long)B * N * D;
@@ -36,9 +50,11 @@ void edge_msg_in_fwd_kernel(const float* __restrict__ states, bf16_t* __restrict
     const float* send = states + ((long)b * V + sender_node(i, d, N, R)) * S;
     bf16_t out[64];
     float p2 = 1e-6f;
-    float e[16];
+    float e[16], er[16], es_[16];
+    edge_state_of(recv, er, S, mode);
+    edge_state_of(send, es_, S, mode);
     for (int s = 0; s < S; ++s) {
-      e[s] = recv[s] - send[s];
+      e[s] = er[s] - es_[s];
       if (s < pdim) p2 += e[s] * e[s];
     }
     const float n = sqrtf(p2);
@@ -59,11 +75,14 @@ void edge_msg_in_fwd_kernel(const float* __restrict__ states, bf16_t* __restrict
 }
 
 // vjp of one slot's edge features wrt the raw diff v (recomputed forward)
-__device__ __forceinline__ void slot_vjp(const float* recv, const float* send,
+__device__ __forceinline__ void slot_vjp(const float* recv_es, const float* send_raw,
                                          const bf16_t* dx, float* acc, float sign,
-                                         int S, int pdim, float comm) {
+                                         int S, int pdim, float comm, int mode) {
   float p2 = 1e-6f;
-  float e[16], g[16];
+  float e[16], g[16], se[16];
+  edge_state_of(send_raw, se, S, mode);
+  const float* recv = recv_es;
+  const float* send = se;
   // vector-load the cotangent row (dx rows are 8-element aligned: KP % 32 == 0)
   for (int s = 0; s < S; s += 8) {
     const bf16x8 v = *(const bf16x8*)(dx + s);
@@ -91,7 +110,7 @@ __device__ __forceinline__ void slot_vjp(const float* recv, const float* send,
 __launch_bounds__(256) __global__
 void edge_msg_in_bwd_kernel(const float* __restrict__ states, const bf16_t* __restrict__ dX,
                             float* __restrict__ dstates, int B, int N, int R, int S,
-                            int pdim, int KP, float comm) {
+                            int pdim, int KP, float comm, int mode) {
   const int D = N + 1 + R;
   const int V = 2 * N + N * R;
   const long total = (long)B * V;
@@ -103,30 +122,44 @@ void edge_msg_in_bwd_kernel(const float* __restrict__ states, const bf16_t* __re
     const bf16_t* dxb = dX + (long)b * N * D * KP;
     float acc[16];
     for (int s = 0; s < S; ++s) acc[s] = 0.f;
+    const float* own = st + (long)v * S;
+    float own_es[16];
+    edge_state_of(own, own_es, S, mode);
     if (v < N) {
       // agent j: receiver side over its D slots, sender side in others' rows
       const int j = v;
-      const float* recv = st + (long)j * S;
       for (int d = 0; d < D; ++d) {
         const float* send = st + (long)sender_node(j, d, N, R) * S;
-        slot_vjp(recv, send, dxb + ((long)j * D + d) * KP, acc, 1.f, S, pdim, comm);
+        slot_vjp(own_es, send, dxb + ((long)j * D + d) * KP, acc, 1.f, S, pdim, comm, mode);
       }
       for (int i = 0; i < N; ++i) {
-        const float* r2 = st + (long)i * S;
-        slot_vjp(r2, recv, dxb + ((long)i * D + j) * KP, acc, -1.f, S, pdim, comm);
+        float r2[16];
+        edge_state_of(st + (long)i * S, r2, S, mode);
+        slot_vjp(r2, own, dxb + ((long)i * D + j) * KP, acc, -1.f, S, pdim, comm, mode);
       }
     } else if (v < 2 * N) {
       const int j = v - N;  // goal j: sender in slot (j, N)
-      slot_vjp(st + (long)j * S, st + (long)v * S, dxb + ((long)j * D + N) * KP, acc, -1.f,
-               S, pdim, comm);
+      float rj[16];
+      edge_state_of(st + (long)j * S, rj, S, mode);
+      slot_vjp(rj, own, dxb + ((long)j * D + N) * KP, acc, -1.f, S, pdim, comm, mode);
     } else {
       const int h = v - 2 * N;  // lidar hit (j, r): sender in slot (j, N+1+r)
       const int j = h / R;
       const int r = h % R;
-      slot_vjp(st + (long)j * S, st + (long)v * S, dxb + ((long)j * D + N + 1 + r) * KP, acc,
-               -1.f, S, pdim, comm);
+      float rj[16];
+      edge_state_of(st + (long)j * S, rj, S, mode);
+      slot_vjp(rj, own, dxb + ((long)j * D + N + 1 + r) * KP, acc, -1.f, S, pdim, comm, mode);
     }
     float* out = dstates + ((long)b * V + v) * S;
-    for (int s = 0; s < S; ++s) out[s] = acc[s];
+    if (mode == 1) {
+      // chain through d(es)/d(state): [x, y, v cos, v sin]
+      const float th = own[2], vv = own[3];
+      out[0] = acc[0];
+      out[1] = acc[1];
+      out[2] = acc[2] * (-vv * __sinf(th)) + acc[3] * (vv * __cosf(th));
+      out[3] = acc[2] * __cosf(th) + acc[3] * __sinf(th);
+    } else {
+      for (int s = 0; s < S; ++s) out[s] = acc[s];
+    }
   }
 }

@@ -428,3 +428,31 @@ def test_fused_env_step_matches_torch():
     assert ((hits_f - hits_r).abs() * close).max() < 1e-3
     mism = (res_fused.graph.mask != ref_graph.mask).float().mean()
     assert mism < 1e-3, mism  # boundary ties only
+
+
+def test_edge_msg_in_dubins_mode():
+    """Transform-mode (DubinsCar) fused edge input vs CPU compose + autograd."""
+    from gcbfplus_amd.env import make_env
+
+    torch.manual_seed(15)
+    env = make_env("DubinsCar", num_agents=4, area_size=2.0, max_step=4, device="cpu")
+    g = env.reset(2, np.random.default_rng(15))
+    st = g.states.clone()
+    st[:, :4, 3] += 0.3  # nonzero speeds exercise the transform
+    mi_cpu = ops.edge_msg_in(st, 4, env.n_rays, 2, env.params["comm_radius"], mode=1)
+    mi_gpu = ops.edge_msg_in(st.cuda(), 4, env.n_rays, 2, env.params["comm_radius"],
+                             mode=1).float().cpu()
+    assert (mi_gpu - mi_cpu).abs().max() < 5e-3
+
+    stg = st.cuda().requires_grad_(True)
+    mi = ops.edge_msg_in(stg, 4, env.n_rays, 2, env.params["comm_radius"], mode=1)
+    gout = torch.randn_like(mi.float()).to(torch.bfloat16)
+    mi.backward(gout)
+    stc = st.clone().requires_grad_(True)
+    mi_c = ops.edge_msg_in(stc, 4, env.n_rays, 2, env.params["comm_radius"], mode=1)
+    mi_c.backward(gout.float().cpu())
+    # only agent rows feed autograd in the training loss
+    ga = stg.grad[:, :4].cpu()
+    gc = stc.grad[:, :4]
+    denom = gc.abs().mean().clamp_min(1e-4)
+    assert ((ga - gc).abs().mean() / denom) < 3e-2, (ga - gc).abs().max()
