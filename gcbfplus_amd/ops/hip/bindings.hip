@@ -12,6 +12,8 @@ template <int ACT>
 __global__ void gemm_bias_act_sm_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 template <int ACT>
 __global__ void gemm_bias_act_bn128_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
+template <int ACT>
+__global__ void gemm_bias_act_glds_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 __global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int);
 template <int ACT, typename OutT>
 __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, OutT*, int, int, int);
@@ -101,7 +103,7 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
     if (act == 0) launch(gemm_bias_act_sm_kernel<0>);
     else if (act == 1) launch(gemm_bias_act_sm_kernel<1>);
     else launch(gemm_bias_act_sm_kernel<2>);
-  } else if (N >= 128 && (N % 128) == 0 && getenv("GCBF_GEMM_BN64") == nullptr) {
+  } else if (N >= 128 && (N % 128) == 0 && getenv("GCBF_GEMM_BN128") != nullptr) {
     // BN=128: halves A re-reads for the 256-wide layers
     dim3 grid((M + 127) / 128, N / 128);
     size_t smem = (size_t)(K / 8) * 128 * 8 * sizeof(uint16_t) + 128 * 40 * sizeof(uint16_t);
@@ -112,6 +114,17 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
     if (act == 0) launch(gemm_bias_act_bn128_kernel<0>);
     else if (act == 1) launch(gemm_bias_act_bn128_kernel<1>);
     else launch(gemm_bias_act_bn128_kernel<2>);
+  } else if (M % 128 == 0 && K % 64 == 0 && getenv("GCBF_GEMM_NOGLDS") == nullptr) {
+    // glds-pipelined path (double-buffered direct-to-LDS A staging)
+    dim3 grid(M / 128, (N + 63) / 64);
+    size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 2 * 128 * 64 * sizeof(uint16_t);
+    auto launch = [&](auto kernel) {
+      hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(x), bfp(w),
+                         bias.data_ptr<float>(), bfp_mut(y), (int)M, (int)N, (int)K);
+    };
+    if (act == 0) launch(gemm_bias_act_glds_kernel<0>);
+    else if (act == 1) launch(gemm_bias_act_glds_kernel<1>);
+    else launch(gemm_bias_act_glds_kernel<2>);
   } else {
     dim3 grid((M + 127) / 128, (N + 63) / 64);
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 128 * 40 * sizeof(uint16_t);

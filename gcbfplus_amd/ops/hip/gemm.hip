@@ -515,3 +515,114 @@ template __global__ void gemv_bias_act_kernel<2, bf16_t>(const bf16_t*, const bf
 template __global__ void gemv_bias_act_kernel<0, float>(const bf16_t*, const bf16_t*, const float*, float*, int, int, int);
 template __global__ void gemv_bias_act_kernel<1, float>(const bf16_t*, const bf16_t*, const float*, float*, int, int, int);
 template __global__ void gemv_bias_act_kernel<2, float>(const bf16_t*, const bf16_t*, const float*, float*, int, int, int);
+
+// ---------------------------------------------------------------------------
+// glds-pipelined main GEMM: BK=64, double-buffered A staged by
+// global_load_lds (16 B per lane), XOR-swizzled LDS image so the b128
+// fragment reads stay bank-conflict-free (granule g' = g ^ ((row>>1)&7) on
+// the [128][64] bf16 tile). Requires M % 128 == 0 and K % 64 == 0 (glds has
+// no per-lane predication); the launcher falls back otherwise.
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void gemm_bias_act_glds_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
+                               const float* __restrict__ bias, bf16_t* __restrict__ Y,
+                               int M, int N, int K) {
+  constexpr int BM = 128, BN = 64, BK = 64;
+  extern __shared__ char smem[];
+  bf16_t* sB = (bf16_t*)smem;                                    // [K/8][BN][8]
+  bf16_t* sA = (bf16_t*)(smem + (K / 8) * BN * 8 * sizeof(bf16_t));  // [2][128][64]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wm = w >> 1, wn = w & 1;  // wave tile: 64(M) x 32(N)
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  // B preload (same image as the base kernel)
+  for (int c = tid; c < K * 8; c += 256) {
+    const int k = c >> 3;
+    const int co = (c & 7) * 8;
+    bf16_t v[8];
+    if (n0 + co + 7 < N) {
+      *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
+  }
+
+  // glds stage of one A tile into buffer `buf`: 16 wave-chunks of 1 KiB
+  auto stage_A = [&](int k0, int buf) {
+    bf16_t* dstbase = sA + buf * 128 * 64;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = (it * 4 + w) * 64 + lane;  // 16-byte granule index
+      const int row = idx >> 3;
+      const int g = idx & 7;
+      const int gsrc = g ^ ((row >> 1) & 7);     // pre-swizzled source
+      const bf16_t* src = X + (long)(m0 + row) * K + k0 + gsrc * 8;
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) uint32_t*)src,
+                                       (__attribute__((address_space(3))) uint32_t*)(dstbase + (long)(it * 4 + w) * 512),
+                                       16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  stage_A(0, 0);
+  const int ntiles = K / BK;
+  for (int t = 0; t < ntiles; ++t) {
+    __syncthreads();  // drains the in-flight glds for buffer t&1
+    if (t + 1 < ntiles) stage_A((t + 1) * BK, (t + 1) & 1);
+    const bf16_t* buf = sA + (t & 1) * 128 * 64;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {  // two 16x16x32 K-steps per tile
+      bf16x8 afr[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        const int row = wm * 64 + mf * 16 + (lane & 15);
+        const int g = kk * 4 + (lane >> 4);
+        const int gs = g ^ ((row >> 1) & 7);
+        afr[mf] = *(const bf16x8*)(buf + (long)row * 64 + gs * 8);
+      }
+      bf16x8 bfr[2];
+      const int kb = t * BK + kk * 32;
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        bfr[nf] = *(const bf16x8*)(sB + (((kb >> 3) + (lane >> 4)) * BN + wn * 32 + nf * 16 + (lane & 15)) * 8);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[mf], bfr[nf], acc[mf][nf], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      const int col = n0 + wn * 32 + nf * 16 + (lane & 15);
+      if (col >= N) continue;
+      const float bv = bias[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = m0 + wm * 64 + mf * 16 + (lane >> 4) * 4 + r;
+        Y[row * N + col] = (bf16_t)apply_act(acc[mf][nf][r] + bv, ACT);
+      }
+    }
+  }
+}
+
+template __global__ void gemm_bias_act_glds_kernel<0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_glds_kernel<1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_glds_kernel<2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
