@@ -156,8 +156,8 @@ std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
   long M = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == M);
   long gk = (K + 63) / 64, gn = (N + 63) / 64;
-  // deterministic split count: aim for ~512 blocks, depends on shapes only
-  long S = std::min<long>(64, std::max<long>(1, 512 / std::max<long>(1, gk * gn)));
+  // deterministic split count: aim for ~1024 blocks, depends on shapes only
+  long S = std::min<long>(128, std::max<long>(1, 1024 / std::max<long>(1, gk * gn)));
   S = std::min<long>(S, std::max<long>(1, (M + 31) / 32));
 
   auto opts = x.options().dtype(torch::kFloat32);
