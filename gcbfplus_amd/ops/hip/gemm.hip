@@ -334,7 +334,9 @@ __launch_bounds__(256) __global__
 void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
                             float* __restrict__ partial, float* __restrict__ db_partial,
                             int M, int N, int K, int S) {
-  constexpr int BKDIM = 64, BN = 64, BMR = 32, TPAD = 40;
+  // 64x64 output tile, BMR=64 reduction steps with register-prefetch
+  // staging (load tile t+1 into registers while tile t computes).
+  constexpr int BKDIM = 64, BN = 64, BMR = 64, TPAD = 72;  // pad >= BMR + 8
   __shared__ bf16_t sXT[BKDIM][TPAD];   // [k][m] transposed X tile
   __shared__ bf16_t sB[BMR / 8][BN][8]; // dZ tile, m-blocked
   __shared__ float sDb[4][BN];          // db tree reduce (k0==0 blocks)
@@ -346,7 +348,6 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
   const int n0 = blockIdx.y * BN;
   const int s = blockIdx.z;
 
-  // split-M range for this s (fixed boundaries -> deterministic)
   const long m_per = ((long)M + S - 1) / S;
   const long ms = (long)s * m_per;
   const long me = (ms + m_per < (long)M) ? ms + m_per : (long)M;
@@ -356,64 +357,84 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
   for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
-  float db_acc = 0.f;  // column sum of dZ (blockIdx.x == 0 only)
+  float db_acc = 0.f;
   const int db_c = tid & 63, db_q = tid >> 6;
 
-  for (long m0 = ms; m0 < me; m0 += BMR) {
-    __syncthreads();
-    {  // stage X^T: thread t loads X[m0 + t>>3][k0 + (t&7)*8 ..+8]
-      const long mr = m0 + (tid >> 3);
-      const int kc = (tid & 7) * 8;
-      bf16_t v[8];
+  // per-thread staging assignments (2 chunks each for X and dZ)
+  //   X: chunk c -> row m0 + (c>>3), k-chunk (c&7)*8 (64 rows x 8 chunks)
+  //   dZ: chunk c -> row m0 + (c>>3), n-chunk (c&7)*8
+  bf16_t xv[2][8], zv[2][8];
+
+  auto load_tile = [&](long m0) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int c = tid + h * 256;
+      const long mr = m0 + (c >> 3);
+      const int kc = (c & 7) * 8;
       if (mr < me && k0 + kc + 7 < K) {
-        *(bf16x8*)v = *(const bf16x8*)(X + mr * K + k0 + kc);
+        *(bf16x8*)xv[h] = *(const bf16x8*)(X + mr * K + k0 + kc);
       } else {
 #pragma unroll
         for (int i = 0; i < 8; ++i)
-          v[i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
+          xv[h][i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
       }
+      if (mr < me && n0 + kc + 7 < N) {
+        *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + mr * N + n0 + kc);
+      } else {
 #pragma unroll
-      for (int i = 0; i < 8; ++i) sXT[kc + i][tid >> 3] = v[i];
-    }
-    {  // stage dZ m-blocked: chunk c -> mrow = c>>3, col8 = (c&7)*8
-      for (int c = tid; c < BMR * 8; c += 256) {
-        const long mr = m0 + (c >> 3);
-        const int co = (c & 7) * 8;
-        bf16_t v[8];
-        if (mr < me && n0 + co + 7 < N) {
-          *(bf16x8*)v = *(const bf16x8*)(dZ + mr * N + n0 + co);
-        } else {
-#pragma unroll
-          for (int i = 0; i < 8; ++i)
-            v[i] = (mr < me && n0 + co + i < N) ? dZ[mr * N + n0 + co + i] : (bf16_t)0.f;
-        }
-#pragma unroll
-        for (int i = 0; i < 8; ++i) sB[(c >> 3) >> 3][co + i][(c >> 3) & 7] = v[i];
+        for (int i = 0; i < 8; ++i)
+          zv[h][i] = (mr < me && n0 + kc + i < N) ? dZ[mr * N + n0 + kc + i] : (bf16_t)0.f;
       }
     }
+  };
+
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int c = tid + h * 256;
+      const int mr = c >> 3;
+      const int kc = (c & 7) * 8;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) sXT[kc + i][mr] = xv[h][i];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) sB[mr >> 3][kc + i][mr & 7] = zv[h][i];
+    }
+  };
+
+  load_tile(ms);
+  for (long m0 = ms; m0 < me; m0 += BMR) {
     __syncthreads();
+    write_tile();
+    __syncthreads();
+    if (m0 + BMR < me) load_tile(m0 + BMR);  // overlap with the MFMA below
 
     if (blockIdx.x == 0) {
-      // free db: the dZ tile is already staged; rows q*8..q*8+7 of col c
-      const bf16x8 v = *(const bf16x8*)(&sB[db_q][db_c][0]);
 #pragma unroll
-      for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
+      for (int q = 0; q < 2; ++q) {
+        const bf16x8 v = *(const bf16x8*)(&sB[db_q * 2 + q][db_c][0]);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
+      }
     }
-    bf16x8 afr[2], bfr[2];
 #pragma unroll
-    for (int kf = 0; kf < 2; ++kf)
-      afr[kf] = *(const bf16x8*)(&sXT[wk * 32 + kf * 16 + (lane & 15)][(lane >> 4) * 8]);
+    for (int mm = 0; mm < 2; ++mm) {  // two 32-deep reduction steps
+      bf16x8 afr[2], bfr[2];
 #pragma unroll
-    for (int nf = 0; nf < 2; ++nf)
-      bfr[nf] = *(const bf16x8*)(&sB[lane >> 4][wn * 32 + nf * 16 + (lane & 15)][0]);
-#pragma unroll
-    for (int kf = 0; kf < 2; ++kf)
+      for (int kf = 0; kf < 2; ++kf)
+        afr[kf] = *(const bf16x8*)(&sXT[wk * 32 + kf * 16 + (lane & 15)][mm * 32 + (lane >> 4) * 8]);
 #pragma unroll
       for (int nf = 0; nf < 2; ++nf)
-        acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
+        bfr[nf] = *(const bf16x8*)(&sB[mm * 4 + (lane >> 4)][wn * 32 + nf * 16 + (lane & 15)][0]);
+#pragma unroll
+      for (int kf = 0; kf < 2; ++kf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
+    }
   }
 
   if (blockIdx.x == 0) {
+    // db_acc covers rows {db_q*16..+16} interleaved (q pairs): tree-reduce
     sDb[db_q][db_c] = db_acc;
     __syncthreads();
     if (db_q == 0 && n0 + db_c < N)
