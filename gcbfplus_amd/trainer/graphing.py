@@ -41,6 +41,7 @@ class GraphedMinibatchStep:
         self.algo = algo
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.fb: Optional[FlatBatch] = None
+        self.full = False
 
     def _alloc(self, batch: FlatBatch):
         mb = self.algo.batch_size
@@ -60,6 +61,9 @@ class GraphedMinibatchStep:
         algo.actor_optim.gflat.zero_()
         total, _ = algo._loss(self.fb, want_info=False)
         total.backward()
+        if self.full:  # single-GPU: optimizer inside the capture too
+            algo.cbf_optim.step()
+            algo.actor_optim.step()
 
     def run(self, batch: FlatBatch, idx: torch.Tensor) -> bool:
         """Returns True if the graphed path handled this minibatch."""
@@ -82,15 +86,17 @@ class GraphedMinibatchStep:
         torch.index_select(batch.unsafe, 0, idx, out=fb.unsafe)
         if fb.u_qp is not None:
             torch.index_select(batch.u_qp, 0, idx, out=fb.u_qp)
+        from ..parallel import dp
+
         if self.graph is None:
+            self.full = not dp.is_active()
             self.graph = _capture(self._body)
         # stream capture records without executing -> always replay
         self.graph.replay()
-        from ..parallel import dp
-
-        dp.allreduce_mean_flat([algo.cbf_optim.gflat, algo.actor_optim.gflat])
-        algo.cbf_optim.step()
-        algo.actor_optim.step()
+        if not self.full:  # DP: all-reduce between backward and the step
+            dp.allreduce_mean_flat([algo.cbf_optim.gflat, algo.actor_optim.gflat])
+            algo.cbf_optim.step()
+            algo.actor_optim.step()
         return True
 
 
