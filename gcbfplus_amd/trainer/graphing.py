@@ -90,7 +90,16 @@ class GraphedMinibatchStep:
 
         if self.graph is None:
             self.full = not dp.is_active()
+            # capture warmup EXECUTES the body (incl. optimizer steps on the
+            # full path): snapshot the optimizer/param state and restore it
+            # so capture is side-effect free; the replay below applies the
+            # one real update.
+            snap = []
+            for opt in (algo.cbf_optim, algo.actor_optim):
+                snap.append({k: v.clone() for k, v in opt.state_dict().items()})
             self.graph = _capture(self._body)
+            for opt, sd in zip((algo.cbf_optim, algo.actor_optim), snap):
+                opt.load_state_dict(sd)
         # stream capture records without executing -> always replay
         self.graph.replay()
         if not self.full:  # DP: all-reduce between backward and the step
