@@ -25,6 +25,15 @@ from torch import Tensor
 
 _EXT = None
 _EXT_ERR: Optional[str] = None
+_ZBIAS = {}
+
+
+def _zero_bias(n: int, device) -> "Tensor":
+    key = (n, str(device))
+    z = _ZBIAS.get(key)
+    if z is None:
+        z = _ZBIAS[key] = torch.zeros(n, device=device)
+    return z
 
 ACT_NONE, ACT_RELU, ACT_TANH = 0, 1, 2
 
@@ -82,7 +91,7 @@ class _FusedLinearHIP(torch.autograd.Function):
         ext = _require_ext()
         x_bf = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
         w_bf = w.to(torch.bfloat16)
-        bias = b if b is not None else torch.zeros(w.shape[1], device=w.device)
+        bias = b if b is not None else _zero_bias(w.shape[1], w.device)
         y = ext.gemm_bias_act(x_bf.contiguous(), w_bf.contiguous(), bias.contiguous(), act)
         ctx.save_for_backward(x_bf, w_bf, y)
         ctx.act = act
@@ -108,7 +117,7 @@ class _FusedLinearHIP(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             # dx = dz @ w^T : feed the row-major GEMM with w^T (small, cheap)
             wt = w_bf.t().contiguous()
-            dx = ext.gemm_bias_act(dz, wt, torch.zeros(wt.shape[1], device=wt.device), ACT_NONE)
+            dx = ext.gemm_bias_act(dz, wt, _zero_bias(wt.shape[1], wt.device), ACT_NONE)
             dx = dx.to(ctx.x_dtype)
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[2]:
             dw, db = ext.gemm_tn(x_bf, dz)  # f32 (K,N), (N,)

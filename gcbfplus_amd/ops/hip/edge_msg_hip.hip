@@ -80,16 +80,13 @@ __device__ __forceinline__ void slot_vjp(const float* recv_es, const float* send
                                          int S, int pdim, float comm, int mode) {
   float p2 = 1e-6f;
   float e[16], g[16], se[16];
-  edge_state_of(send_raw, se, S, mode);
   const float* recv = recv_es;
-  const float* send = se;
-  // vector-load the cotangent row (dx rows are 8-element aligned: KP % 32 == 0)
-  for (int s = 0; s < S; s += 8) {
-    const bf16x8 v = *(const bf16x8*)(dx + s);
-#pragma unroll
-    for (int i = 0; i < 8; ++i)
-      if (s + i < 16) g[s + i] = (float)v[i];
+  const float* send = send_raw;
+  if (mode != 0) {
+    edge_state_of(send_raw, se, S, mode);
+    send = se;
   }
+  for (int s = 0; s < S; ++s) g[s] = (float)dx[s];
   for (int s = 0; s < S; ++s) {
     e[s] = recv[s] - send[s];
     if (s < pdim) p2 += e[s] * e[s];
@@ -123,8 +120,12 @@ void edge_msg_in_bwd_kernel(const float* __restrict__ states, const bf16_t* __re
     float acc[16];
     for (int s = 0; s < S; ++s) acc[s] = 0.f;
     const float* own = st + (long)v * S;
-    float own_es[16];
-    edge_state_of(own, own_es, S, mode);
+    float own_es_buf[16];
+    const float* own_es = own;
+    if (mode != 0) {
+      edge_state_of(own, own_es_buf, S, mode);
+      own_es = own_es_buf;
+    }
     if (v < N) {
       // agent j: receiver side over its D slots, sender side in others' rows
       const int j = v;
@@ -133,21 +134,27 @@ void edge_msg_in_bwd_kernel(const float* __restrict__ states, const bf16_t* __re
         slot_vjp(own_es, send, dxb + ((long)j * D + d) * KP, acc, 1.f, S, pdim, comm, mode);
       }
       for (int i = 0; i < N; ++i) {
+        const float* ri = st + (long)i * S;
         float r2[16];
-        edge_state_of(st + (long)i * S, r2, S, mode);
-        slot_vjp(r2, own, dxb + ((long)i * D + j) * KP, acc, -1.f, S, pdim, comm, mode);
+        if (mode != 0) {
+          edge_state_of(ri, r2, S, mode);
+          ri = r2;
+        }
+        slot_vjp(ri, own, dxb + ((long)i * D + j) * KP, acc, -1.f, S, pdim, comm, mode);
       }
     } else if (v < 2 * N) {
       const int j = v - N;  // goal j: sender in slot (j, N)
-      float rj[16];
-      edge_state_of(st + (long)j * S, rj, S, mode);
+      const float* rj = st + (long)j * S;
+      float rb[16];
+      if (mode != 0) { edge_state_of(rj, rb, S, mode); rj = rb; }
       slot_vjp(rj, own, dxb + ((long)j * D + N) * KP, acc, -1.f, S, pdim, comm, mode);
     } else {
       const int h = v - 2 * N;  // lidar hit (j, r): sender in slot (j, N+1+r)
       const int j = h / R;
       const int r = h % R;
-      float rj[16];
-      edge_state_of(st + (long)j * S, rj, S, mode);
+      const float* rj = st + (long)j * S;
+      float rb[16];
+      if (mode != 0) { edge_state_of(rj, rb, S, mode); rj = rb; }
       slot_vjp(rj, own, dxb + ((long)j * D + N + 1 + r) * KP, acc, -1.f, S, pdim, comm, mode);
     }
     float* out = dstates + ((long)b * V + v) * S;
