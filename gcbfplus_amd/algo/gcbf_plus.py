@@ -196,14 +196,22 @@ class GCBFPlus(GCBF):
         safe_m = mb.safe.reshape(-1)
         unsafe_m = mb.unsafe.reshape(-1)
 
-        # action = 2*actor + u_ref (the deployed policy)
-        action = 2 * self.actor(g, e, msg_in=mi) + env.u_ref(g)
-        next_g = env.forward_graph(g, action)
-
-        # h and h_next in ONE batched CBF forward (2B graphs): halves the
-        # GEMM call count and doubles M for better CU fill
+        # action = 2*actor + u_ref (the deployed policy); h and h_next in ONE
+        # batched CBF forward (2B graphs): halves the GEMM call count and
+        # doubles M for better CU fill
+        raw = self.actor(g, e, msg_in=mi)
+        prep = env.loss_prep(g, raw) if hasattr(env, "loss_prep") else None
+        if prep is not None:
+            # fused K16 kernel: u_ref + action clamp + euler + [cur; next]
+            action, big_states = prep
+            next_g = GraphBatch(states=big_states[B:], mask=g.mask,
+                                n_agents=g.n_agents, n_rays=g.n_rays)
+        else:
+            action = 2 * raw + env.u_ref(g)
+            next_g = env.forward_graph(g, action)
+            big_states = torch.cat([g.states, next_g.states])
         big = GraphBatch(
-            states=torch.cat([g.states, next_g.states]),
+            states=big_states,
             mask=torch.cat([g.mask, g.mask]),
             n_agents=g.n_agents, n_rays=g.n_rays,
         )

@@ -273,6 +273,22 @@ class DoubleIntegrator(MultiAgentEnv):
         next_agent = self.agent_step_euler(graph.agent_states, action)
         return graph.with_agent_states(next_agent)
 
+    def loss_prep(self, graph: GraphBatch, raw: Tensor):
+        """Fused GCBF+ loss prologue (K16): u_ref -> action = clamp(2*raw +
+        u_ref) -> euler next state -> [states; next_states]. Only for the
+        exact class (subclasses override dynamics/u_ref)."""
+        if type(self) is not DoubleIntegrator or not graph.states.is_cuda:
+            return None
+        from .. import ops
+
+        if not ops.hip_available():
+            return None
+        p = self._params
+        return ops.di_loss_prep(
+            graph.states, raw, self._K.to(graph.device), self.num_agents,
+            self._dt, 1.0 / p["m"], p["comm_radius"], 0.5,
+        )
+
     # ---- cost / reward (reference :183-198) --------------------------------
     def get_cost(self, graph: GraphBatch) -> Tensor:
         pos = graph.agent_states[..., : self.pos_dim]

@@ -35,6 +35,11 @@ __global__ void proxqp_kernel(const float*, const float*, const float*, const fl
 __global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float, int);
 __global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float, int);
 __global__ void gcbf_loss_fwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, float*, long, int, float, float, float, float, float, float, float);
+__global__ void di_loss_prep_fwd_kernel(const float*, const float*, const float*, float*, float*,
+                                        int, int, int, float, float, float, float);
+__global__ void di_loss_prep_bwd_kernel(const float*, const float*, const float*, const float*,
+                                        const float*, const float*, float*, int, int, int,
+                                        float, float, float, float);
 __global__ void di_env_step_kernel(const float*, const float*, const float*, const float*,
                                    float*, bool*, float*, float*, int, int, int, float,
                                    float, float, float, float);
@@ -381,7 +386,42 @@ std::vector<torch::Tensor> di_env_step(torch::Tensor states, torch::Tensor actio
   return {nxt, mask, reward, cost};
 }
 
+std::vector<torch::Tensor> di_loss_prep_fwd(torch::Tensor states, torch::Tensor raw,
+                                            torch::Tensor Kmat, long N, double dt,
+                                            double inv_m, double comm, double vmax) {
+  CHECK_IN(states);
+  CHECK_IN(raw);
+  long B = states.size(0), V = states.size(1);
+  auto action = torch::empty({B, N, 2}, states.options());
+  auto big = torch::empty({2 * B, V, 4}, states.options());
+  long rows = 2 * B * V;
+  hipLaunchKernelGGL(di_loss_prep_fwd_kernel, dim3((rows + 255) / 256), dim3(256), 0,
+                     cur_stream(), states.data_ptr<float>(), raw.data_ptr<float>(),
+                     Kmat.data_ptr<float>(), action.data_ptr<float>(), big.data_ptr<float>(),
+                     (int)B, (int)V, (int)N, (float)dt, (float)inv_m, (float)comm,
+                     (float)vmax);
+  return {action, big};
+}
+
+torch::Tensor di_loss_prep_bwd(torch::Tensor states, torch::Tensor raw, torch::Tensor Kmat,
+                               torch::Tensor action, torch::Tensor daction,
+                               torch::Tensor dbig, long N, double dt, double inv_m,
+                               double comm, double vmax) {
+  long B = states.size(0), V = states.size(1);
+  auto draw = torch::empty_like(raw);
+  long total = B * N;
+  hipLaunchKernelGGL(di_loss_prep_bwd_kernel, dim3((total + 255) / 256), dim3(256), 0,
+                     cur_stream(), states.data_ptr<float>(), raw.data_ptr<float>(),
+                     Kmat.data_ptr<float>(), action.data_ptr<float>(),
+                     daction.data_ptr<float>(), dbig.data_ptr<float>(),
+                     draw.data_ptr<float>(), (int)B, (int)V, (int)N, (float)dt,
+                     (float)inv_m, (float)comm, (float)vmax);
+  return draw;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("di_loss_prep_fwd", &di_loss_prep_fwd);
+  m.def("di_loss_prep_bwd", &di_loss_prep_bwd);
   m.def("di_env_step", &di_env_step, "fused DoubleIntegrator env step (K5-K8)");
   m.def("gcbf_loss_fwd", &gcbf_loss_fwd);
   m.def("gcbf_loss_bwd", &gcbf_loss_bwd);

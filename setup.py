@@ -21,6 +21,7 @@ setup(
                 "gcbfplus_amd/ops/hip/edge_msg.hip",
                 "gcbfplus_amd/ops/hip/loss.hip",
                 "gcbfplus_amd/ops/hip/env_step.hip",
+                "gcbfplus_amd/ops/hip/loss_prep.hip",
                 "gcbfplus_amd/ops/hip/optimizer.hip",
                 "gcbfplus_amd/ops/hip/bindings.hip",
             ],

@@ -456,3 +456,43 @@ def test_edge_msg_in_dubins_mode():
     gc = stc.grad[:, :4]
     denom = gc.abs().mean().clamp_min(1e-4)
     assert ((ga - gc).abs().mean() / denom) < 3e-2, (ga - gc).abs().max()
+
+
+def test_di_loss_prep_matches_compose():
+    """K16 fused loss prologue vs the composed-torch path, values + grads."""
+    from gcbfplus_amd.env import make_env
+
+    env = make_env("DoubleIntegrator", num_agents=8, area_size=4.0, max_step=8,
+                   device="cuda")
+    rng = np.random.default_rng(31)
+    g = env.reset(6, rng)
+    torch.manual_seed(31)
+    raw = (torch.randn(6, 8, 2, device="cuda") * 0.6).requires_grad_(True)
+    K = env._K.to("cuda")
+    p = env._params
+    act_f, big_f = ops.di_loss_prep(g.states, raw, K, 8, env._dt,
+                                    1.0 / p["m"], p["comm_radius"], 0.5)
+    da = torch.randn_like(act_f)
+    db = torch.randn_like(big_f)
+    (act_f * da).sum().backward(retain_graph=True)
+    g1 = raw.grad.clone()
+    raw.grad = None
+    (big_f * db).sum().backward()
+    g2 = raw.grad.clone()
+
+    # CPU compose path (forced by moving to CPU)
+    act_c, big_c = ops.di_loss_prep(g.states.cpu(), raw.detach().cpu(), K.cpu(), 8,
+                                    env._dt, 1.0 / p["m"], p["comm_radius"], 0.5)
+    assert torch.allclose(act_f.cpu(), act_c, atol=1e-5)
+    assert torch.allclose(big_f.cpu(), big_c, atol=1e-5)
+    # grads: compose backward on CPU against the HIP grads
+    raw3 = raw.detach().cpu().requires_grad_(True)
+    a3, b3 = ops.di_loss_prep(g.states.cpu(), raw3, K.cpu(), 8, env._dt,
+                              1.0 / p["m"], p["comm_radius"], 0.5)
+    (a3 * da.cpu()).sum().backward()
+    assert torch.allclose(g1.cpu(), raw3.grad, atol=1e-5)
+    raw4 = raw.detach().cpu().requires_grad_(True)
+    a4, b4 = ops.di_loss_prep(g.states.cpu(), raw4, K.cpu(), 8, env._dt,
+                              1.0 / p["m"], p["comm_radius"], 0.5)
+    (b4 * db.cpu()).sum().backward()
+    assert torch.allclose(g2.cpu(), raw4.grad, atol=1e-5)
