@@ -67,6 +67,7 @@ class GNNLayer(nn.Module):
         send_idx: Tensor,  # (N, D)
         agents_only: bool,
         msg_in: Tensor = None,  # optional fused (B, N, D, K[pad]) input
+        onehot_nodes: bool = False,  # node_feats ARE the constant one-hots
     ) -> Tensor:
         B, V, F = node_feats.shape
         N, D = mask.shape[1], mask.shape[2]
@@ -78,6 +79,16 @@ class GNNLayer(nn.Module):
         msg = self.msg_out(self.msg_mlp(msg_in))  # (B,N,D,msg_dim)
         gate = self.attn_out(self.attn_mlp(msg)).squeeze(-1)  # (B,N,D)
         aggr = ops.masked_softmax_aggr(gate, msg, mask)  # (B,N,msg_dim)
+        d0 = self.update_mlp.layers[0]
+        if agents_only and onehot_nodes and d0.in_dim == 3 + self.msg_dim:
+            # agent one-hot is [0,0,1]: cat([onehot, aggr]) @ W ==
+            # aggr @ W[3:] + (b + W[2]) — drops the cat/cast AND shrinks the
+            # GEMM K from 3+msg_dim (padded) to msg_dim (glds-aligned)
+            h = ops.fused_linear(aggr, d0.kernel[3:], d0.bias + d0.kernel[2],
+                                 d0.act)
+            for l in self.update_mlp.layers[1:]:
+                h = l(h)
+            return self.update_out(h)
         if agents_only:
             upd_in = torch.cat([node_feats[:, :N], aggr.to(node_feats.dtype)], dim=-1)
         else:
@@ -117,6 +128,7 @@ class GNN(nn.Module):
             device = msg_in0.device
         else:
             device = edge_feats.device
+        onehot = node_feats is None
         if node_feats is None:
             node_feats = one_hot_node_feats(B, N, R, device, torch.float32)
         send_idx = sender_index(N, R, device)
@@ -124,7 +136,8 @@ class GNN(nn.Module):
         for i, layer in enumerate(self.layers):
             last = i == n_layers - 1
             x = layer(x, edge_feats, graph.mask, send_idx, agents_only=last,
-                      msg_in=msg_in0 if i == 0 else None)
+                      msg_in=msg_in0 if i == 0 else None,
+                      onehot_nodes=onehot and i == 0)
         return x  # (B, N, out_dim)
 
 
