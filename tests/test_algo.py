@@ -150,3 +150,51 @@ def test_value_net_and_tanh_normal():
     assert torch.isfinite(lp).all()
     m = pol.mode(g, e)
     assert m.shape == a.shape
+
+
+def test_tanh_transformed_distribution():
+    """TanhTransformedDistribution (reference distribution.py:10-60):
+    log_prob matches the change-of-variables formula inside the threshold,
+    returns tail log-mass at the boundary, samples stay in (-1, 1)."""
+    import math
+
+    from gcbfplus_amd.algo.module import TanhTransformedDistribution
+
+    torch.manual_seed(3)
+    loc = torch.randn(32) * 0.5
+    scale = torch.rand(32) * 0.5 + 0.2
+    d = TanhTransformedDistribution(loc, scale)
+    s = d.sample()
+    assert s.abs().max() < 1.0
+    v = torch.tanh(torch.randn(32) * 0.5)
+    lp = d.log_prob(v)
+    base = torch.distributions.Normal(loc, scale)
+    ref = base.log_prob(torch.atanh(v)) - torch.log1p(-v * v)
+    assert torch.allclose(lp, ref, atol=1e-5)
+    # boundary: tail mass, finite
+    lp_edge = d.log_prob(torch.ones(32))
+    ref_edge = (1 - base.cdf(torch.full((32,), math.atanh(0.999)))).clamp_min(1e-38).log()
+    assert torch.allclose(lp_edge, ref_edge, atol=1e-4)
+    assert torch.isfinite(d.entropy()).all()
+    assert torch.allclose(d.mode(), torch.tanh(loc))
+
+
+def test_nn_utils_helpers():
+    """nn/utils parity helpers (reference nn/utils.py:19-51)."""
+    from gcbfplus_amd.nn.utils import (default_nn_init, get_act_from_str, safe_get,
+                                       scaled_init, signal_last_enumerate)
+
+    w = torch.empty(64, 32)
+    default_nn_init(w)
+    lim = (6.0 / (64 + 32)) ** 0.5
+    assert w.abs().max() <= lim
+    w2 = torch.empty(64, 32)
+    scaled_init(default_nn_init, 0.01)(w2)
+    assert w2.abs().max() <= lim * 0.01
+    assert get_act_from_str("relu") is torch.relu
+    out = list(signal_last_enumerate("abc"))
+    assert out == [(False, 0, "a"), (False, 1, "b"), (True, 2, "c")]
+    arr = torch.arange(12.0).reshape(4, 3)
+    got = safe_get(arr, torch.tensor([0, 3, 4, -1]))
+    assert torch.allclose(got[:2], arr[[0, 3]])
+    assert torch.isnan(got[2]).all() and torch.isnan(got[3]).all()
