@@ -14,10 +14,11 @@ template <int ACT>
 __global__ void gemm_bias_act_bn128_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 template <int ACT>
 __global__ void gemm_bias_act_glds_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
+__global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int);
 template <int ACT, typename OutT>
 __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, OutT*, int, int, int);
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
-__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, float*, float*, float*, float*, int*, int, int, int, int);
+__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
@@ -165,21 +166,15 @@ std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
   auto opts = x.options().dtype(torch::kFloat32);
   auto partial = torch::empty({S, K, N}, opts);
   auto db_partial = torch::empty({S, N}, opts);
+  auto stream = cur_stream();
+  hipLaunchKernelGGL(gemm_tn_partial_kernel, dim3(gk, gn, S), dim3(256), 0, stream,
+                     bfp(x), bfp(dz), partial.data_ptr<float>(), db_partial.data_ptr<float>(),
+                     (int)M, (int)N, (int)K, (int)S);
   auto dw = torch::empty({K, N}, opts);
   auto db = torch::empty({N}, opts);
-  // persistent zeroed counters for the fused last-block reduce (each group
-  // resets its slot to 0, so one allocation serves every launch — and stays
-  // HIP-graph-capture safe after first use)
-  static torch::Tensor counters;
-  long ng = gk * gn;
-  if (!counters.defined() || counters.numel() < ng ||
-      counters.device() != x.device())
-    counters = torch::zeros({std::max<long>(ng, 4096)},
-                            x.options().dtype(torch::kInt32));
-  hipLaunchKernelGGL(gemm_tn_partial_kernel, dim3(gk, gn, S), dim3(256), 0, cur_stream(),
-                     bfp(x), bfp(dz), partial.data_ptr<float>(), db_partial.data_ptr<float>(),
-                     dw.data_ptr<float>(), db.data_ptr<float>(),
-                     counters.data_ptr<int>(), (int)M, (int)N, (int)K, (int)S);
+  hipLaunchKernelGGL(reduce_dw_db_kernel, dim3((K * N + N + 255) / 256), dim3(256), 0, stream,
+                     partial.data_ptr<float>(), db_partial.data_ptr<float>(),
+                     dw.data_ptr<float>(), db.data_ptr<float>(), K * N, (int)N, (int)S);
   return {dw, db};
 }
 

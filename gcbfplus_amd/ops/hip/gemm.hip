@@ -333,8 +333,7 @@ __global__ void act_bwd_kernel(const bf16_t* __restrict__ dY, const bf16_t* __re
 __launch_bounds__(256) __global__
 void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
                             float* __restrict__ partial, float* __restrict__ db_partial,
-                            float* __restrict__ dW, float* __restrict__ db,
-                            int* __restrict__ counters, int M, int N, int K, int S) {
+                            int M, int N, int K, int S) {
   // 64x64 output tile, BMR=64 reduction steps with register-prefetch
   // staging (load tile t+1 into registers while tile t computes).
   constexpr int BKDIM = 64, BN = 64, BMR = 64, TPAD = 72;  // pad >= BMR + 8
@@ -455,45 +454,31 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
         if (krow < K) out[(long)krow * N + col] = acc[kf][nf][r];
       }
     }
+}
 
-  // ---- fused stage 2: LAST block of each (gk, gn) group reduces the S
-  // partial slices of its own 64x64 tile (fixed s-order: deterministic).
-  // The persistent counter returns to 0 afterwards so the buffer needs no
-  // per-launch memset.
-  __threadfence();
-  __shared__ int s_grp;
-  if (tid == 0) {
-    const int g = blockIdx.y * gridDim.x + blockIdx.x;
-    s_grp = (atomicAdd(&counters[g], 1) == S - 1) ? g : -1;
+// (dw_partial (S,K,N), db_partial (S,N)) -> (dW, db) in ONE launch
+// (fixed-order sums: deterministic; 4 accumulators hide add latency).
+__global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* __restrict__ pb,
+                                    float* __restrict__ dW, float* __restrict__ db,
+                                    long KN, int N, int S) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const bool is_db = i >= KN;
+  if (i >= KN + N) return;
+  const float* src = is_db ? pb : pw;
+  const long stride = is_db ? N : KN;
+  const long off = is_db ? i - KN : i;
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int s = 0;
+  for (; s + 4 <= S; s += 4) {
+    a0 += src[(long)s * stride + off];
+    a1 += src[(long)(s + 1) * stride + off];
+    a2 += src[(long)(s + 2) * stride + off];
+    a3 += src[(long)(s + 3) * stride + off];
   }
-  __syncthreads();
-  if (s_grp < 0) return;
-  __threadfence();
-  const long KN = (long)K * N;
-  for (int e = tid; e < BKDIM * BN; e += 256) {
-    const int kr = k0 + e / BN, c = n0 + e % BN;
-    if (kr >= K || c >= N) continue;
-    const long off = (long)kr * N + c;
-    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-    int ss = 0;
-    for (; ss + 4 <= S; ss += 4) {
-      a0 += partial[(long)ss * KN + off];
-      a1 += partial[(long)(ss + 1) * KN + off];
-      a2 += partial[(long)(ss + 2) * KN + off];
-      a3 += partial[(long)(ss + 3) * KN + off];
-    }
-    for (; ss < S; ++ss) a0 += partial[(long)ss * KN + off];
-    dW[off] = (a0 + a1) + (a2 + a3);
-  }
-  if (blockIdx.x == 0) {
-    for (int c = tid; c < BN; c += 256) {
-      if (n0 + c >= N) continue;
-      float a = 0.f;
-      for (int ss = 0; ss < S; ++ss) a += db_partial[(long)ss * N + n0 + c];
-      db[n0 + c] = a;
-    }
-  }
-  if (tid == 0) counters[s_grp] = 0;
+  for (; s < S; ++s) a0 += src[(long)s * stride + off];
+  const float r = (a0 + a1) + (a2 + a3);
+  if (is_db) db[off] = r;
+  else dW[off] = r;
 }
 
 // ---------------------------------------------------------------------------
