@@ -14,7 +14,7 @@ template <int ACT>
 __global__ void gemm_bias_act_bn128_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 template <int ACT>
 __global__ void gemm_bias_act_glds_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
-__global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int);
+__global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int, int);
 template <int ACT, typename OutT>
 __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, OutT*, int, int, int);
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
@@ -31,6 +31,7 @@ template <int MAXNV, int MAXK>
 __global__ void proxqp_kernel(const float*, const float*, const float*, const float*, const float*,
                               const float*, float*, int, int, int, int, float, float, float);
 __global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float, int);
+__global__ void edge_msg_in_fwd_s4_kernel(const float*, bf16_t_*, int, int, int, float);
 __global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float, int);
 __global__ void gcbf_loss_fwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, float*, long, int, float, float, float, float, float, float, float);
 __global__ void di_loss_prep_fwd_kernel(const float*, const float*, const float*, float*, float*,
@@ -153,7 +154,8 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, long act) {
   return dz;
 }
 
-std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
+std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
+                                        torch::Tensor dw, torch::Tensor db, bool acc) {
   CHECK_IN(x);
   CHECK_IN(dz);
   long M = x.size(0), K = x.size(1), N = dz.size(1);
@@ -170,12 +172,24 @@ std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
   hipLaunchKernelGGL(gemm_tn_partial_kernel, dim3(gk, gn, S), dim3(256), 0, stream,
                      bfp(x), bfp(dz), partial.data_ptr<float>(), db_partial.data_ptr<float>(),
                      (int)M, (int)N, (int)K, (int)S);
-  auto dw = torch::empty({K, N}, opts);
-  auto db = torch::empty({N}, opts);
   hipLaunchKernelGGL(reduce_dw_db_kernel, dim3((K * N + N + 255) / 256), dim3(256), 0, stream,
                      partial.data_ptr<float>(), db_partial.data_ptr<float>(),
-                     dw.data_ptr<float>(), db.data_ptr<float>(), K * N, (int)N, (int)S);
+                     dw.data_ptr<float>(), db.data_ptr<float>(), K * N, (int)N, (int)S,
+                     acc ? 1 : 0);
   return {dw, db};
+}
+
+std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto dw = torch::empty({x.size(1), dz.size(1)}, opts);
+  auto db = torch::empty({dz.size(1)}, opts);
+  return gemm_tn_impl(x, dz, dw, db, false);
+}
+
+void gemm_tn_acc(torch::Tensor x, torch::Tensor dz, torch::Tensor dw, torch::Tensor db) {
+  TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() && db.is_contiguous());
+  TORCH_CHECK(dw.size(0) == x.size(1) && dw.size(1) == dz.size(1) && db.size(0) == dz.size(1));
+  gemm_tn_impl(x, dz, dw, db, true);
 }
 
 std::vector<torch::Tensor> softmax_aggr_fwd(torch::Tensor gate, torch::Tensor msg,
@@ -293,6 +307,12 @@ torch::Tensor edge_msg_in_fwd(torch::Tensor states, long N, long R, long pdim, l
   TORCH_CHECK(S <= 16 && KP <= 64);
   auto X = torch::empty({B, N, D, KP}, states.options().dtype(torch::kBFloat16));
   long total = B * N * D;
+  if (mode == 0 && S == 4 && pdim == 2 && KP == 32) {
+    hipLaunchKernelGGL(edge_msg_in_fwd_s4_kernel, dim3((total + 255) / 256), dim3(256), 0,
+                       cur_stream(), states.data_ptr<float>(), bfp_mut(X), (int)B, (int)N,
+                       (int)R, (float)comm);
+    return X;
+  }
   hipLaunchKernelGGL(edge_msg_in_fwd_kernel, dim3((total + 255) / 256), dim3(256), 0,
                      cur_stream(), states.data_ptr<float>(), bfp_mut(X), (int)B, (int)N,
                      (int)R, (int)S, (int)pdim, (int)KP, (float)comm, (int)mode);
@@ -431,6 +451,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bias_act", &gemm_bias_act, "Y = act(X@W + b), MFMA bf16");
   m.def("act_bwd", &act_bwd, "dZ = dY * act'(Y)");
   m.def("gemm_tn", &gemm_tn, "dW = X^T dZ, db = colsum dZ (deterministic)");
+  m.def("gemm_tn_acc", &gemm_tn_acc, "gemm_tn accumulated (+=) into given f32 buffers");
   m.def("softmax_aggr_fwd", &softmax_aggr_fwd);
   m.def("softmax_aggr_bwd", &softmax_aggr_bwd);
   m.def("raytrace_rect", &raytrace_rect);

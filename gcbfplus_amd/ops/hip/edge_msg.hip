@@ -73,6 +73,50 @@ void edge_msg_in_fwd_kernel(const float* __restrict__ states, bf16_t* __restrict
   }
 }
 
+// S=4 / mode 0 / pdim 2 / KP 32 specialization (DoubleIntegrator, the
+// benchmark env): float4 state loads instead of 8 scalar dwords per node —
+// the generic kernel's runtime-S copy loops serialize into scalar loads
+// that dominate at ~3 waves/SIMD occupancy.
+__launch_bounds__(256) __global__
+void edge_msg_in_fwd_s4_kernel(const float* __restrict__ states, bf16_t* __restrict__ X,
+                               int B, int N, int R, float comm) {
+  const int D = N + 1 + R;
+  const int V = 2 * N + N * R;
+  const long total = (long)B * N * D;
+  for (long row = (long)blockIdx.x * blockDim.x + threadIdx.x; row < total;
+       row += (long)gridDim.x * blockDim.x) {
+    const int d = row % D;
+    const int i = (row / D) % N;
+    const int b = row / ((long)N * D);
+    const float4 rv = *(const float4*)(states + ((long)b * V + i) * 4);
+    const float4 sv = *(const float4*)(states + ((long)b * V + sender_node(i, d, N, R)) * 4);
+    const float e0 = rv.x - sv.x, e1 = rv.y - sv.y;
+    const float e2 = rv.z - sv.z, e3 = rv.w - sv.w;
+    const float n = sqrtf(1e-6f + e0 * e0 + e1 * e1);
+    const float coef = (n > comm) ? comm / n : 1.f;
+    const int stype = (d < N) ? 0 : (d == N ? 1 : 2);
+    bf16_t out[16];
+    out[0] = (bf16_t)(e0 * coef);
+    out[1] = (bf16_t)(e1 * coef);
+    out[2] = (bf16_t)e2;
+    out[3] = (bf16_t)e3;
+    out[4] = (bf16_t)(stype == 2 ? 1.f : 0.f);
+    out[5] = (bf16_t)(stype == 1 ? 1.f : 0.f);
+    out[6] = (bf16_t)(stype == 0 ? 1.f : 0.f);
+    out[7] = (bf16_t)0.f;
+    out[8] = (bf16_t)0.f;
+    out[9] = (bf16_t)1.f;
+#pragma unroll
+    for (int s = 10; s < 16; ++s) out[s] = (bf16_t)0.f;
+    bf16_t* dst = X + row * 32;
+    *(bf16x8*)dst = *(bf16x8*)out;
+    *(bf16x8*)(dst + 8) = *(bf16x8*)(out + 8);
+    const bf16x8 z8 = {};
+    *(bf16x8*)(dst + 16) = z8;
+    *(bf16x8*)(dst + 24) = z8;
+  }
+}
+
 // vjp of one slot's edge features wrt the raw diff v (recomputed forward)
 __device__ __forceinline__ void slot_vjp(const float* recv_es, const float* send_raw,
                                          const bf16_t* dx, float* acc, float sign,

@@ -458,9 +458,11 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
 
 // (dw_partial (S,K,N), db_partial (S,N)) -> (dW, db) in ONE launch
 // (fixed-order sums: deterministic; 4 accumulators hide add latency).
+// acc != 0 accumulates (+=) into dW/db — used to write straight into the
+// optimizer's flat-gradient views, skipping autograd's AccumulateGrad adds.
 __global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* __restrict__ pb,
                                     float* __restrict__ dW, float* __restrict__ db,
-                                    long KN, int N, int S) {
+                                    long KN, int N, int S, int acc) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const bool is_db = i >= KN;
   if (i >= KN + N) return;
@@ -477,8 +479,8 @@ __global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* _
   }
   for (; s < S; ++s) a0 += src[(long)s * stride + off];
   const float r = (a0 + a1) + (a2 + a3);
-  if (is_db) db[off] = r;
-  else dW[off] = r;
+  float* dst = is_db ? db + off : dW + off;
+  *dst = acc ? (*dst + r) : r;
 }
 
 // ---------------------------------------------------------------------------

@@ -496,3 +496,28 @@ def test_di_loss_prep_matches_compose():
                               1.0 / p["m"], p["comm_radius"], 0.5)
     (b4 * db.cpu()).sum().backward()
     assert torch.allclose(g2.cpu(), raw4.grad, atol=1e-5)
+
+
+def test_fused_linear_direct_grad_accumulate():
+    """When params carry live .grad buffers (FusedAdamW flat views), backward
+    writes += into them and returns None; results must match the standard
+    AccumulateGrad path bitwise."""
+    torch.manual_seed(40)
+    M, K, N = 512, 64, 128
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, N, device="cuda", requires_grad=True)
+    b = torch.randn(N, device="cuda", requires_grad=True)
+    y = ops.fused_linear(x, w, b, ops.ACT_RELU)
+    g = torch.randn_like(y)
+    (y * g).sum().backward()
+    dw_ref, db_ref = w.grad.clone(), b.grad.clone()
+
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    base = torch.full((K, N), 3.0, device="cuda")
+    w2.grad = base.clone()  # pre-existing buffer: direct += path
+    b2.grad = torch.full((N,), 5.0, device="cuda")
+    y2 = ops.fused_linear(x, w2, b2, ops.ACT_RELU)
+    (y2 * g).sum().backward()
+    assert torch.equal(w2.grad - 3.0, dw_ref)
+    assert torch.equal(b2.grad - 5.0, db_ref)
