@@ -32,6 +32,7 @@ __global__ void proxqp_kernel(const float*, const float*, const float*, const fl
                               const float*, float*, int, int, int, int, float, float, float);
 __global__ void edge_msg_in_fwd_kernel(const float*, bf16_t_*, int, int, int, int, int, int, float, int);
 __global__ void edge_msg_in_fwd_s4_kernel(const float*, bf16_t_*, int, int, int, float);
+__global__ void edge_msg_in_bwd_s4_kernel(const float*, const bf16_t_*, float*, int, int, int, float);
 __global__ void edge_msg_in_bwd_kernel(const float*, const bf16_t_*, float*, int, int, int, int, int, int, float, int);
 __global__ void gcbf_loss_fwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, float*, long, int, float, float, float, float, float, float, float);
 __global__ void di_loss_prep_fwd_kernel(const float*, const float*, const float*, float*, float*,
@@ -327,6 +328,12 @@ torch::Tensor edge_msg_in_bwd(torch::Tensor states, torch::Tensor dX, long N, lo
   long KP = dX.size(-1);
   auto dstates = torch::empty_like(states);
   long total = B * V;
+  if (mode == 0 && S == 4 && pdim == 2 && KP == 32) {
+    hipLaunchKernelGGL(edge_msg_in_bwd_s4_kernel, dim3((total + 255) / 256), dim3(256), 0,
+                       cur_stream(), states.data_ptr<float>(), bfp(dX),
+                       dstates.data_ptr<float>(), (int)B, (int)N, (int)R, (float)comm);
+    return dstates;
+  }
   hipLaunchKernelGGL(edge_msg_in_bwd_kernel, dim3((total + 255) / 256), dim3(256), 0,
                      cur_stream(), states.data_ptr<float>(), bfp(dX),
                      dstates.data_ptr<float>(), (int)B, (int)N, (int)R, (int)S, (int)pdim,

@@ -117,6 +117,74 @@ void edge_msg_in_fwd_s4_kernel(const float* __restrict__ states, bf16_t* __restr
   }
 }
 
+// S=4 / mode 0 / pdim 2 / KP 32 backward specialization: float4 state loads,
+// one bf16x8 load per dX row (only the first 4 lanes carry gradient — the
+// one-hot / pad columns are constants).
+__device__ __forceinline__ void slot_vjp_s4(const float4 rv, const float4 sv,
+                                            const bf16_t* __restrict__ dx,
+                                            float* __restrict__ acc, float sign,
+                                            float comm) {
+  const bf16x8 gv = *(const bf16x8*)dx;
+  const float g0 = (float)gv[0], g1 = (float)gv[1];
+  const float g2 = (float)gv[2], g3 = (float)gv[3];
+  const float e0 = rv.x - sv.x, e1 = rv.y - sv.y;
+  const float n = sqrtf(1e-6f + e0 * e0 + e1 * e1);
+  if (n > comm) {
+    const float inv_n = 1.f / n;
+    const float gdotp = g0 * e0 + g1 * e1;
+    const float c3 = gdotp * inv_n * inv_n * inv_n;
+    acc[0] += sign * comm * (g0 * inv_n - e0 * c3);
+    acc[1] += sign * comm * (g1 * inv_n - e1 * c3);
+  } else {
+    acc[0] += sign * g0;
+    acc[1] += sign * g1;
+  }
+  acc[2] += sign * g2;
+  acc[3] += sign * g3;
+}
+
+__launch_bounds__(256) __global__
+void edge_msg_in_bwd_s4_kernel(const float* __restrict__ states,
+                               const bf16_t* __restrict__ dX,
+                               float* __restrict__ dstates, int B, int N, int R,
+                               float comm) {
+  const int D = N + 1 + R;
+  const int V = 2 * N + N * R;
+  const long total = (long)B * V;
+  for (long node = (long)blockIdx.x * blockDim.x + threadIdx.x; node < total;
+       node += (long)gridDim.x * blockDim.x) {
+    const int v = node % V;
+    const int b = node / V;
+    const float* st = states + (long)b * V * 4;
+    const bf16_t* dxb = dX + (long)b * N * D * 32;
+    const float4 own = *(const float4*)(st + (long)v * 4);
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    if (v < N) {
+      const int j = v;
+      for (int d = 0; d < D; ++d) {
+        const float4 sv = *(const float4*)(st + (long)sender_node(j, d, N, R) * 4);
+        slot_vjp_s4(own, sv, dxb + ((long)j * D + d) * 32, acc, 1.f, comm);
+      }
+      for (int i = 0; i < N; ++i) {
+        const float4 rv = *(const float4*)(st + (long)i * 4);
+        slot_vjp_s4(rv, own, dxb + ((long)i * D + j) * 32, acc, -1.f, comm);
+      }
+    } else if (v < 2 * N) {
+      const int j = v - N;
+      const float4 rv = *(const float4*)(st + (long)j * 4);
+      slot_vjp_s4(rv, own, dxb + ((long)j * D + N) * 32, acc, -1.f, comm);
+    } else {
+      const int h = v - 2 * N;
+      const int j = h / R;
+      const int r = h % R;
+      const float4 rv = *(const float4*)(st + (long)j * 4);
+      slot_vjp_s4(rv, own, dxb + ((long)j * D + N + 1 + r) * 32, acc, -1.f, comm);
+    }
+    *(float4*)(dstates + ((long)b * V + v) * 4) =
+        float4{acc[0], acc[1], acc[2], acc[3]};
+  }
+}
+
 // vjp of one slot's edge features wrt the raw diff v (recomputed forward)
 __device__ __forceinline__ void slot_vjp(const float* recv_es, const float* send_raw,
                                          const bf16_t* dx, float* acc, float sign,

@@ -514,10 +514,14 @@ def test_fused_linear_direct_grad_accumulate():
 
     w2 = w.detach().clone().requires_grad_(True)
     b2 = b.detach().clone().requires_grad_(True)
-    base = torch.full((K, N), 3.0, device="cuda")
-    w2.grad = base.clone()  # pre-existing buffer: direct += path
-    b2.grad = torch.full((N,), 5.0, device="cuda")
+    w2.grad = torch.zeros(K, N, device="cuda")  # pre-existing buffer: += path
+    b2.grad = torch.zeros(N, device="cuda")
     y2 = ops.fused_linear(x, w2, b2, ops.ACT_RELU)
     (y2 * g).sum().backward()
-    assert torch.equal(w2.grad - 3.0, dw_ref)
-    assert torch.equal(b2.grad - 5.0, db_ref)
+    assert torch.equal(w2.grad, dw_ref)
+    assert torch.equal(b2.grad, db_ref)
+    # accumulation onto non-zero content (fp add, not overwrite)
+    y3 = ops.fused_linear(x, w2, b2, ops.ACT_RELU)
+    (y3 * g).sum().backward()
+    assert torch.allclose(w2.grad, 2 * dw_ref, rtol=1e-6, atol=1e-6)
+    assert torch.allclose(b2.grad, 2 * db_ref, rtol=1e-6, atol=1e-5)
