@@ -95,9 +95,12 @@ void softmax_aggr_bwd_kernel(const bf16_t* __restrict__ daggr, const float* __re
 
   for (int d = tid; d < D; d += 256) dgate[row * D + d] = at[d] * (sv[d] - t);
 
-  // dmsg[d][c] = attn[d] * daggr[c]
-  for (long i = tid; i < (long)D * C; i += 256) {
-    const int d = i / C, c = i % C;
-    dmsg[row * (long)D * C + i] = (bf16_t)(at[d] * (float)da[c]);
+  // dmsg[d][c] = attn[d] * daggr[c] — wave per d (coalesced lanes, no
+  // per-element integer division)
+  bf16_t* dm = dmsg + row * (long)D * C;
+  for (int d = w; d < D; d += 4) {
+    const float a = at[d];
+    for (int c = lane; c < C; c += WAVE)
+      dm[(long)d * C + c] = (bf16_t)(a * (float)da[c]);
   }
 }
