@@ -77,6 +77,14 @@ class GCBF(MultiAgentController):
         self.rng = np.random.default_rng(seed=seed + 1 + 7919 * dp.rank())
         self._mb_graph = None
         dp.broadcast_modules([self.cbf, self.actor])
+        self._refresh_optim_bf16()
+
+    def _refresh_optim_bf16(self):
+        """Re-sync FusedAdamW bf16 weight shadows after out-of-band param
+        writes (checkpoint load / DP broadcast)."""
+        for opt in (getattr(self, "cbf_optim", None), getattr(self, "actor_optim", None)):
+            if opt is not None and hasattr(opt, "refresh_bf16"):
+                opt.refresh_bf16()
 
     def _graphed_mb(self):
         if self._mb_graph is None:
@@ -133,6 +141,7 @@ class GCBF(MultiAgentController):
             net_from_flax_tree(self.actor, pickle.load(f), "PolicyHead", "OutputDense")
         with open(os.path.join(path, "cbf.pkl"), "rb") as f:
             net_from_flax_tree(self.cbf, pickle.load(f), "CBFHead", "Dense_0")
+        self._refresh_optim_bf16()
 
     # ---- full training resume (NOT in the reference, which saves params
     # only — gcbf.py:344-357 / SURVEY §5.4) --------------------------------

@@ -36,6 +36,8 @@ class GCBFPlus(GCBF):
         self.cbf_tgt = copy.deepcopy(self.cbf)
         for p in self.cbf_tgt.parameters():
             p.requires_grad_(False)
+            if hasattr(p, "_bf"):  # never share the live net's bf16 shadow
+                del p._bf
         self.qp_iters = 150
         self.qp_relax_penalty = 1e3
 

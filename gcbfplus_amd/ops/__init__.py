@@ -90,7 +90,11 @@ class _FusedLinearHIP(torch.autograd.Function):
     def forward(ctx, x: Tensor, w: Tensor, b: Optional[Tensor], act: int):
         ext = _require_ext()
         x_bf = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
-        w_bf = w.to(torch.bfloat16)
+        # FusedAdamW maintains a bf16 shadow per param (p._bf, updated inside
+        # the optimizer kernel) — skip the per-call cast when present
+        w_bf = getattr(w, "_bf", None)
+        if w_bf is None:
+            w_bf = w.to(torch.bfloat16)
         bias = b if b is not None else _zero_bias(w.shape[1], w.device)
         y = ext.gemm_bias_act(x_bf.contiguous(), w_bf.contiguous(), bias.contiguous(), act)
         ctx.save_for_backward(x_bf, w_bf, y)
@@ -124,9 +128,14 @@ class _FusedLinearHIP(torch.autograd.Function):
             dz = ext.act_bwd(dy.contiguous().to(torch.bfloat16), y, ctx.act)  # (M,N) bf16
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            # dx = dz @ w^T : feed the row-major GEMM with w^T (small, cheap)
-            wt = w_bf.t().contiguous()
-            dx = ext.gemm_bias_act(dz, wt, _zero_bias(wt.shape[1], wt.device), ACT_NONE)
+            if dz.shape[1] % 32 == 0:
+                # dx = dz @ w^T via the transposed B-stage kernel: no
+                # materialized w^T copy per backward
+                dx = ext.gemm_bt(dz, w_bf)
+            else:
+                # small-N heads: pad path with an explicit transpose
+                wt = w_bf.t().contiguous()
+                dx = ext.gemm_bias_act(dz, wt, _zero_bias(wt.shape[1], wt.device), ACT_NONE)
             dx = dx.to(ctx.x_dtype)
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[2]:
             if ctx.acc is not None:

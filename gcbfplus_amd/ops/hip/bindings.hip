@@ -6,13 +6,13 @@
 
 // kernel decls (defined in the .hip TUs)
 typedef __bf16 bf16_t_;
-template <int ACT>
+template <int ACT, bool BT>
 __global__ void gemm_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
-template <int ACT>
+template <int ACT, bool BT>
 __global__ void gemm_bias_act_sm_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 template <int ACT>
 __global__ void gemm_bias_act_bn128_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
-template <int ACT>
+template <int ACT, bool BT>
 __global__ void gemm_bias_act_glds_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 __global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int, int);
 template <int ACT, typename OutT>
@@ -24,7 +24,7 @@ __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
 __global__ void grad_norm_sq_partial_kernel(const float*, long, float*);
 __global__ void reduce_norm_kernel(const float*, int, float*);
-__global__ void adamw_flat_kernel(float*, const float*, float*, float*, const float*, const int*,
+__global__ void adamw_flat_kernel(float*, const float*, float*, float*, bf16_t_*, const float*, const int*,
                                   float, float, float, float, float, float, long);
 __global__ void advance_step_kernel(int*, const float*);
 template <int MAXNV, int MAXK>
@@ -105,9 +105,9 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
       hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(x), bfp(w),
                          bias.data_ptr<float>(), bfp_mut(y), (int)M, (int)N, (int)K);
     };
-    if (act == 0) launch(gemm_bias_act_sm_kernel<0>);
-    else if (act == 1) launch(gemm_bias_act_sm_kernel<1>);
-    else launch(gemm_bias_act_sm_kernel<2>);
+    if (act == 0) launch(gemm_bias_act_sm_kernel<0, false>);
+    else if (act == 1) launch(gemm_bias_act_sm_kernel<1, false>);
+    else launch(gemm_bias_act_sm_kernel<2, false>);
   } else if (N >= 128 && (N % 128) == 0 && getenv("GCBF_GEMM_BN128") != nullptr) {
     // BN=128: halves A re-reads for the 256-wide layers
     dim3 grid((M + 127) / 128, N / 128);
@@ -127,9 +127,9 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
       hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(x), bfp(w),
                          bias.data_ptr<float>(), bfp_mut(y), (int)M, (int)N, (int)K);
     };
-    if (act == 0) launch(gemm_bias_act_glds_kernel<0>);
-    else if (act == 1) launch(gemm_bias_act_glds_kernel<1>);
-    else launch(gemm_bias_act_glds_kernel<2>);
+    if (act == 0) launch(gemm_bias_act_glds_kernel<0, false>);
+    else if (act == 1) launch(gemm_bias_act_glds_kernel<1, false>);
+    else launch(gemm_bias_act_glds_kernel<2, false>);
   } else {
     dim3 grid((M + 127) / 128, (N + 63) / 64);
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 128 * 40 * sizeof(uint16_t);
@@ -137,9 +137,9 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
       hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(x), bfp(w),
                          bias.data_ptr<float>(), bfp_mut(y), (int)M, (int)N, (int)K);
     };
-    if (act == 0) launch(gemm_bias_act_kernel<0>);
-    else if (act == 1) launch(gemm_bias_act_kernel<1>);
-    else launch(gemm_bias_act_kernel<2>);
+    if (act == 0) launch(gemm_bias_act_kernel<0, false>);
+    else if (act == 1) launch(gemm_bias_act_kernel<1, false>);
+    else launch(gemm_bias_act_kernel<2, false>);
   }
   return y;
 }
@@ -191,6 +191,40 @@ void gemm_tn_acc(torch::Tensor x, torch::Tensor dz, torch::Tensor dw, torch::Ten
   TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() && db.is_contiguous());
   TORCH_CHECK(dw.size(0) == x.size(1) && dw.size(1) == dz.size(1) && db.size(0) == dz.size(1));
   gemm_tn_impl(x, dz, dw, db, true);
+}
+
+
+// dX = dZ @ W^T with W the original forward weight (N, K): no transpose copy.
+torch::Tensor gemm_bt(torch::Tensor dz, torch::Tensor w) {
+  CHECK_IN(dz);
+  CHECK_IN(w);
+  TORCH_CHECK(dz.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
+  long M = dz.size(0), K = dz.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "W cols must equal dZ cols");
+  TORCH_CHECK(K % 32 == 0, "BT path needs reduction dim % 32 == 0");
+  auto y = torch::empty({M, N}, dz.options());
+  auto stream = cur_stream();
+  static torch::Tensor zb;  // zero bias cache (per device lifetime)
+  if (!zb.defined() || zb.numel() < N || zb.device() != dz.device())
+    zb = torch::zeros({std::max<long>(N, 512)}, dz.options().dtype(torch::kFloat32));
+  const float* bias = zb.data_ptr<float>();
+  if (M <= 16384) {
+    dim3 grid((M + 31) / 32, (N + 63) / 64);
+    size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 32 * 40 * sizeof(uint16_t);
+    hipLaunchKernelGGL((gemm_bias_act_sm_kernel<0, true>), grid, dim3(256), smem, stream,
+                       bfp(dz), bfp(w), bias, bfp_mut(y), (int)M, (int)N, (int)K);
+  } else if (M % 128 == 0 && K % 64 == 0 && getenv("GCBF_GEMM_NOGLDS") == nullptr) {
+    dim3 grid(M / 128, (N + 63) / 64);
+    size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 2 * 128 * 64 * sizeof(uint16_t);
+    hipLaunchKernelGGL((gemm_bias_act_glds_kernel<0, true>), grid, dim3(256), smem, stream,
+                       bfp(dz), bfp(w), bias, bfp_mut(y), (int)M, (int)N, (int)K);
+  } else {
+    dim3 grid((M + 127) / 128, (N + 63) / 64);
+    size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 128 * 40 * sizeof(uint16_t);
+    hipLaunchKernelGGL((gemm_bias_act_kernel<0, true>), grid, dim3(256), smem, stream,
+                       bfp(dz), bfp(w), bias, bfp_mut(y), (int)M, (int)N, (int)K);
+  }
+  return y;
 }
 
 std::vector<torch::Tensor> softmax_aggr_fwd(torch::Tensor gate, torch::Tensor msg,
@@ -245,8 +279,9 @@ torch::Tensor raytrace_rect(torch::Tensor pos, torch::Tensor points, long n_rays
 }
 
 torch::Tensor fused_adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-                               torch::Tensor v, torch::Tensor t, double lr, double b1,
-                               double b2, double eps, double wd, double max_norm) {
+                               torch::Tensor v, torch::Tensor pbf, torch::Tensor t,
+                               double lr, double b1, double b2, double eps, double wd,
+                               double max_norm) {
   CHECK_IN(p);
   CHECK_IN(g);
   CHECK_IN(m);
@@ -262,9 +297,9 @@ torch::Tensor fused_adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m
                      partial.data_ptr<float>(), nb, norm.data_ptr<float>());
   hipLaunchKernelGGL(adamw_flat_kernel, dim3(512), dim3(256), 0, stream,
                      p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
-                     v.data_ptr<float>(), norm.data_ptr<float>(), t.data_ptr<int>(),
-                     (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
-                     (float)max_norm, n);
+                     v.data_ptr<float>(), bfp_mut(pbf), norm.data_ptr<float>(),
+                     t.data_ptr<int>(), (float)lr, (float)b1, (float)b2, (float)eps,
+                     (float)wd, (float)max_norm, n);
   hipLaunchKernelGGL(advance_step_kernel, dim3(1), dim3(1), 0, stream,
                      t.data_ptr<int>(), norm.data_ptr<float>());
   return norm;
@@ -457,6 +492,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("proxqp_solve", &proxqp_solve_hip, "batched dense QP, one wave per problem (K11)");
   m.def("gemm_bias_act", &gemm_bias_act, "Y = act(X@W + b), MFMA bf16");
   m.def("act_bwd", &act_bwd, "dZ = dY * act'(Y)");
+  m.def("gemm_bt", &gemm_bt, "dX = dZ @ W^T (transposed B-stage, no copy)");
   m.def("gemm_tn", &gemm_tn, "dW = X^T dZ, db = colsum dZ (deterministic)");
   m.def("gemm_tn_acc", &gemm_tn_acc, "gemm_tn accumulated (+=) into given f32 buffers");
   m.def("softmax_aggr_fwd", &softmax_aggr_fwd);

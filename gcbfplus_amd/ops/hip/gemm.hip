@@ -11,13 +11,59 @@
 //                          partial-sum pass + reduction (no atomics).
 #include "common.h"
 
+
+// ---------------------------------------------------------------------------
+// B-operand staging into the k-blocked LDS image sB[[K/8][BN=64][8]].
+// BT=false: W is (K, N) row-major, stage cols n0..n0+63.
+// BT=true:  W is the ORIGINAL forward weight (N_out, K) row-major and the
+//           GEMM consumes W^T — stage rows n0..n0+63 transposed, so the
+//           backward dX = dZ @ W^T needs NO materialized transpose copy.
+// ---------------------------------------------------------------------------
+template <bool BT>
+__device__ __forceinline__ void stage_B_image(const bf16_t* __restrict__ W, bf16_t* sB,
+                                              int N, int K, int n0, int tid) {
+  constexpr int BN = 64;
+  if constexpr (!BT) {
+    for (int c = tid; c < K * 8; c += 256) {
+      const int k = c >> 3;
+      const int co = (c & 7) * 8;
+      bf16_t v[8];
+      if (n0 + co + 7 < N) {
+        *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
+    }
+  } else {
+    const int kch = K >> 3;  // K % 8 == 0 (launcher enforces)
+    for (int c = tid; c < BN * kch; c += 256) {
+      const int r = c / kch;
+      const int kc = (c % kch) * 8;
+      bf16_t v[8];
+      if (n0 + r < N) {
+        *(bf16x8*)v = *(const bf16x8*)(W + (long)(n0 + r) * K + kc);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[i] = (bf16_t)0.f;
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        sB[(((kc + i) >> 3) * BN + r) * 8 + ((kc + i) & 7)] = v[i];
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Y[M,N] = act(X[M,K] @ W[K,N] + bias[N]);  K % 32 == 0 (launcher pads).
 // grid: (ceil(M/128), ceil(N/64)); block: 256 threads (4 waves, 2x2).
 // LDS: B-block [K/8][64][8] (k-blocked so B-fragments are single b128 reads)
 //      + A-tile [128][40] (pad 32->40 kills b128 bank conflicts).
 // ---------------------------------------------------------------------------
-template <int ACT>
+template <int ACT, bool BT>
 __launch_bounds__(256) __global__
 void gemm_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
                           const float* __restrict__ bias, bf16_t* __restrict__ Y,
@@ -36,20 +82,7 @@ void gemm_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict
 
   // ---- preload W block n0..n0+63 into k-blocked LDS image ----------------
   // chunk c -> k = c>>3, col8 = (c&7)*8 ; threads cover K*8 chunks.
-  for (int c = tid; c < K * 8; c += 256) {
-    const int k = c >> 3;
-    const int co = (c & 7) * 8;
-    bf16_t v[8];
-    if (n0 + co + 7 < N) {
-      *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
-    } else {
-#pragma unroll
-      for (int i = 0; i < 8; ++i)
-        v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
-    }
-#pragma unroll
-    for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
-  }
+  stage_B_image<BT>(W, sB, N, K, n0, tid);
 
   f32x4 acc[4][2];
 #pragma unroll
@@ -207,7 +240,7 @@ template __global__ void gemm_bias_act_bn128_kernel<2>(const bf16_t*, const bf16
 // Small-M variant: BM=32 x BN=64 tile (4 waves as 2x2, wave 16x32) so
 // mid-size rows (update/head layers, M ~ 2k) still fill all 256 CUs.
 // ---------------------------------------------------------------------------
-template <int ACT>
+template <int ACT, bool BT>
 __launch_bounds__(256) __global__
 void gemm_bias_act_sm_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
                              const float* __restrict__ bias, bf16_t* __restrict__ Y,
@@ -224,20 +257,7 @@ void gemm_bias_act_sm_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
 
-  for (int c = tid; c < K * 8; c += 256) {
-    const int k = c >> 3;
-    const int co = (c & 7) * 8;
-    bf16_t v[8];
-    if (n0 + co + 7 < N) {
-      *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
-    } else {
-#pragma unroll
-      for (int i = 0; i < 8; ++i)
-        v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
-    }
-#pragma unroll
-    for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
-  }
+  stage_B_image<BT>(W, sB, N, K, n0, tid);
 
   f32x4 acc[1][2];
   acc[0][0] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -281,9 +301,10 @@ void gemm_bias_act_sm_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   }
 }
 
-template __global__ void gemm_bias_act_sm_kernel<0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_sm_kernel<1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_sm_kernel<2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<0, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<1, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<2, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<0, true>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
 
 // ---------------------------------------------------------------------------
 // Small-N path (N <= 16): one wave per output row, W cached in LDS.
@@ -486,9 +507,10 @@ __global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* _
 // ---------------------------------------------------------------------------
 // host-visible launchers (called from bindings.cpp)
 // ---------------------------------------------------------------------------
-template __global__ void gemm_bias_act_kernel<0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_kernel<1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_kernel<2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<0, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<1, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<2, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<0, true>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
 template __global__ void gemv_bias_act_kernel<0, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
 template __global__ void gemv_bias_act_kernel<1, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
 template __global__ void gemv_bias_act_kernel<2, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
@@ -503,7 +525,7 @@ template __global__ void gemv_bias_act_kernel<2, float>(const bf16_t*, const bf1
 // the [128][64] bf16 tile). Requires M % 128 == 0 and K % 64 == 0 (glds has
 // no per-lane predication); the launcher falls back otherwise.
 // ---------------------------------------------------------------------------
-template <int ACT>
+template <int ACT, bool BT>
 __launch_bounds__(256) __global__
 void gemm_bias_act_glds_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
                                const float* __restrict__ bias, bf16_t* __restrict__ Y,
@@ -521,20 +543,7 @@ void gemm_bias_act_glds_kernel(const bf16_t* __restrict__ X, const bf16_t* __res
   const int n0 = blockIdx.y * BN;
 
   // B preload (same image as the base kernel)
-  for (int c = tid; c < K * 8; c += 256) {
-    const int k = c >> 3;
-    const int co = (c & 7) * 8;
-    bf16_t v[8];
-    if (n0 + co + 7 < N) {
-      *(bf16x8*)v = *(const bf16x8*)(W + (long)k * N + n0 + co);
-    } else {
-#pragma unroll
-      for (int i = 0; i < 8; ++i)
-        v[i] = (n0 + co + i < N) ? W[(long)k * N + n0 + co + i] : (bf16_t)0.f;
-    }
-#pragma unroll
-    for (int i = 0; i < 8; ++i) sB[((k >> 3) * BN + (co + i)) * 8 + (k & 7)] = v[i];
-  }
+  stage_B_image<BT>(W, sB, N, K, n0, tid);
 
   // glds stage of one A tile into buffer `buf`: 16 wave-chunks of 1 KiB
   auto stage_A = [&](int k0, int buf) {
@@ -603,6 +612,7 @@ void gemm_bias_act_glds_kernel(const bf16_t* __restrict__ X, const bf16_t* __res
   }
 }
 
-template __global__ void gemm_bias_act_glds_kernel<0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_glds_kernel<1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_glds_kernel<2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_glds_kernel<0, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_glds_kernel<1, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_glds_kernel<2, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_glds_kernel<0, true>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);

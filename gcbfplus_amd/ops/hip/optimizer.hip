@@ -45,8 +45,11 @@ __global__ void reduce_norm_kernel(const float* __restrict__ partial, int nb,
 
 // p, g, m, v: flat f32 buffers of length n. t: device int32 step counter
 // (pre-increment semantics: kernel uses t+1). norm: device scalar ||g||.
+// pbf: bf16 shadow of p kept in lock-step so fused_linear never re-casts
+// weights on the hot path (one cast kernel per Dense per minibatch saved).
 __global__ void adamw_flat_kernel(float* __restrict__ p, const float* __restrict__ g,
                                   float* __restrict__ m, float* __restrict__ v,
+                                  bf16_t* __restrict__ pbf,
                                   const float* __restrict__ norm, const int* __restrict__ t,
                                   float lr, float b1, float b2, float eps, float wd,
                                   float max_norm, long n) {
@@ -64,7 +67,9 @@ __global__ void adamw_flat_kernel(float* __restrict__ p, const float* __restrict
     m[i] = mi;
     v[i] = vi;
     const float upd = (mi / bc1) / (sqrtf(vi / bc2) + eps) + wd * p[i];
-    p[i] -= lr * upd;
+    const float pn = p[i] - lr * upd;
+    p[i] = pn;
+    pbf[i] = (bf16_t)pn;
   }
 }
 

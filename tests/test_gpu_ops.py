@@ -525,3 +525,29 @@ def test_fused_linear_direct_grad_accumulate():
     (y3 * g).sum().backward()
     assert torch.allclose(w2.grad, 2 * dw_ref, rtol=1e-6, atol=1e-6)
     assert torch.allclose(b2.grad, 2 * db_ref, rtol=1e-6, atol=1e-5)
+
+
+def test_bf16_weight_shadow_in_sync():
+    """FusedAdamW's bf16 shadow (p._bf) must track the f32 master through
+    optimizer steps (kernel-updated) and checkpoint loads (refresh)."""
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.algo import make_algo
+    from gcbfplus_amd.trainer.utils import collect_rollout
+
+    torch.manual_seed(7)
+    env = make_env("DoubleIntegrator", num_agents=8, area_size=4.0, max_step=8,
+                   device="cuda")
+    algo = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim,
+                     n_agents=env.num_agents, gnn_layers=1, batch_size=16,
+                     buffer_size=16, horizon=4, inner_epoch=1, seed=0)
+    rng = np.random.default_rng(3)
+    g = env.reset(2, rng)
+    ro = collect_rollout(env, algo.step, g)
+    algo.update(ro, 0)
+    for net in (algo.cbf, algo.actor):
+        for p in net.parameters():
+            assert hasattr(p, "_bf")
+            assert torch.equal(p._bf, p.detach().to(torch.bfloat16)), p.shape
+    for p in algo.cbf_tgt.parameters():
+        assert not hasattr(p, "_bf")
