@@ -124,14 +124,22 @@ class _FusedLinearHIP(torch.autograd.Function):
             else:
                 dz_f = dy
             dz = dz_f.to(torch.bfloat16).contiguous()
+            actin, yact = ACT_NONE, dz
         else:
-            dz = ext.act_bwd(dy.contiguous().to(torch.bfloat16), y, ctx.act)  # (M,N) bf16
+            # activation backward dz = dy * act'(y) is FOLDED into the two
+            # consumer GEMMs' operand stages (same f32 math, same bf16
+            # rounding as a materialized act_bwd pass)
+            dz = dy.contiguous().to(torch.bfloat16)
+            actin, yact = ctx.act, y
+            if actin != ACT_NONE and dz.shape[1] % 32 != 0:
+                dz = ext.act_bwd(dz, y, actin)  # rare: narrow activated layer
+                actin, yact = ACT_NONE, dz
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             if dz.shape[1] % 32 == 0:
                 # dx = dz @ w^T via the transposed B-stage kernel: no
                 # materialized w^T copy per backward
-                dx = ext.gemm_bt(dz, w_bf)
+                dx = ext.gemm_bt(dz, w_bf, yact, actin)
             else:
                 # small-N heads: pad path with an explicit transpose
                 wt = w_bf.t().contiguous()
@@ -140,9 +148,9 @@ class _FusedLinearHIP(torch.autograd.Function):
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[2]:
             if ctx.acc is not None:
                 wp, bp = ctx.acc
-                ext.gemm_tn_acc(x_bf, dz, wp.grad, bp.grad)  # += into views
+                ext.gemm_tn_acc(x_bf, dz, yact, actin, wp.grad, bp.grad)  # += into views
             else:
-                dw, db = ext.gemm_tn(x_bf, dz)  # f32 (K,N), (N,)
+                dw, db = ext.gemm_tn(x_bf, dz, yact, actin)  # f32 (K,N), (N,)
         if not ctx.has_bias:
             db = None
         return dx, dw, db, None

@@ -6,10 +6,10 @@
 
 // kernel decls (defined in the .hip TUs)
 typedef __bf16 bf16_t_;
-template <int ACT, bool BT>
-__global__ void gemm_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
-template <int ACT, bool BT>
-__global__ void gemm_bias_act_sm_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
+template <int ACT, bool BT, int ACTIN>
+__global__ void gemm_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, const bf16_t_*, int, int, int);
+template <int ACT, bool BT, int ACTIN>
+__global__ void gemm_bias_act_sm_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, const bf16_t_*, int, int, int);
 template <int ACT>
 __global__ void gemm_bias_act_bn128_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 template <int ACT, bool BT>
@@ -18,7 +18,8 @@ __global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, 
 template <int ACT, typename OutT>
 __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, OutT*, int, int, int);
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
-__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+template <int ACT>
+__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
@@ -103,11 +104,12 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 32 * 40 * sizeof(uint16_t);
     auto launch = [&](auto kernel) {
       hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(x), bfp(w),
-                         bias.data_ptr<float>(), bfp_mut(y), (int)M, (int)N, (int)K);
+                         bias.data_ptr<float>(), bfp_mut(y), (const bf16_t_*)nullptr,
+                         (int)M, (int)N, (int)K);
     };
-    if (act == 0) launch(gemm_bias_act_sm_kernel<0, false>);
-    else if (act == 1) launch(gemm_bias_act_sm_kernel<1, false>);
-    else launch(gemm_bias_act_sm_kernel<2, false>);
+    if (act == 0) launch(gemm_bias_act_sm_kernel<0, false, 0>);
+    else if (act == 1) launch(gemm_bias_act_sm_kernel<1, false, 0>);
+    else launch(gemm_bias_act_sm_kernel<2, false, 0>);
   } else if (N >= 128 && (N % 128) == 0 && getenv("GCBF_GEMM_BN128") != nullptr) {
     // BN=128: halves A re-reads for the 256-wide layers
     dim3 grid((M + 127) / 128, N / 128);
@@ -135,11 +137,12 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 128 * 40 * sizeof(uint16_t);
     auto launch = [&](auto kernel) {
       hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(x), bfp(w),
-                         bias.data_ptr<float>(), bfp_mut(y), (int)M, (int)N, (int)K);
+                         bias.data_ptr<float>(), bfp_mut(y), (const bf16_t_*)nullptr,
+                         (int)M, (int)N, (int)K);
     };
-    if (act == 0) launch(gemm_bias_act_kernel<0, false>);
-    else if (act == 1) launch(gemm_bias_act_kernel<1, false>);
-    else launch(gemm_bias_act_kernel<2, false>);
+    if (act == 0) launch(gemm_bias_act_kernel<0, false, 0>);
+    else if (act == 1) launch(gemm_bias_act_kernel<1, false, 0>);
+    else launch(gemm_bias_act_kernel<2, false, 0>);
   }
   return y;
 }
@@ -156,6 +159,7 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, long act) {
 }
 
 std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
+                                        torch::Tensor yact, long actin,
                                         torch::Tensor dw, torch::Tensor db, bool acc) {
   CHECK_IN(x);
   CHECK_IN(dz);
@@ -170,9 +174,15 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   auto partial = torch::empty({S, K, N}, opts);
   auto db_partial = torch::empty({S, N}, opts);
   auto stream = cur_stream();
-  hipLaunchKernelGGL(gemm_tn_partial_kernel, dim3(gk, gn, S), dim3(256), 0, stream,
-                     bfp(x), bfp(dz), partial.data_ptr<float>(), db_partial.data_ptr<float>(),
-                     (int)M, (int)N, (int)K, (int)S);
+  const bf16_t_* ya = actin ? bfp(yact) : nullptr;
+  auto launch = [&](auto kernel) {
+    hipLaunchKernelGGL(kernel, dim3(gk, gn, S), dim3(256), 0, stream,
+                       bfp(x), bfp(dz), ya, partial.data_ptr<float>(),
+                       db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
+  };
+  if (actin == 1) launch(gemm_tn_partial_kernel<1>);
+  else if (actin == 2) launch(gemm_tn_partial_kernel<2>);
+  else launch(gemm_tn_partial_kernel<0>);
   hipLaunchKernelGGL(reduce_dw_db_kernel, dim3((K * N + N + 255) / 256), dim3(256), 0, stream,
                      partial.data_ptr<float>(), db_partial.data_ptr<float>(),
                      dw.data_ptr<float>(), db.data_ptr<float>(), K * N, (int)N, (int)S,
@@ -180,22 +190,26 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   return {dw, db};
 }
 
-std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz) {
+std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz, torch::Tensor yact,
+                                   long actin) {
   auto opts = x.options().dtype(torch::kFloat32);
   auto dw = torch::empty({x.size(1), dz.size(1)}, opts);
   auto db = torch::empty({dz.size(1)}, opts);
-  return gemm_tn_impl(x, dz, dw, db, false);
+  return gemm_tn_impl(x, dz, yact, actin, dw, db, false);
 }
 
-void gemm_tn_acc(torch::Tensor x, torch::Tensor dz, torch::Tensor dw, torch::Tensor db) {
+void gemm_tn_acc(torch::Tensor x, torch::Tensor dz, torch::Tensor yact, long actin,
+                 torch::Tensor dw, torch::Tensor db) {
   TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() && db.is_contiguous());
   TORCH_CHECK(dw.size(0) == x.size(1) && dw.size(1) == dz.size(1) && db.size(0) == dz.size(1));
-  gemm_tn_impl(x, dz, dw, db, true);
+  gemm_tn_impl(x, dz, yact, actin, dw, db, true);
 }
 
 
-// dX = dZ @ W^T with W the original forward weight (N, K): no transpose copy.
-torch::Tensor gemm_bt(torch::Tensor dz, torch::Tensor w) {
+// dX = dZ @ W^T with W the original forward weight (N, K): no transpose
+// copy. actin != 0 folds dZ = dY * act'(Y) into the A-operand stage (dz is
+// then dY and yact the saved activation) — kills the act_bwd pass.
+torch::Tensor gemm_bt(torch::Tensor dz, torch::Tensor w, torch::Tensor yact, long actin) {
   CHECK_IN(dz);
   CHECK_IN(w);
   TORCH_CHECK(dz.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
@@ -208,12 +222,19 @@ torch::Tensor gemm_bt(torch::Tensor dz, torch::Tensor w) {
   if (!zb.defined() || zb.numel() < N || zb.device() != dz.device())
     zb = torch::zeros({std::max<long>(N, 512)}, dz.options().dtype(torch::kFloat32));
   const float* bias = zb.data_ptr<float>();
+  const bf16_t_* ya = actin ? bfp(yact) : nullptr;
   if (M <= 16384) {
     dim3 grid((M + 31) / 32, (N + 63) / 64);
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 32 * 40 * sizeof(uint16_t);
-    hipLaunchKernelGGL((gemm_bias_act_sm_kernel<0, true>), grid, dim3(256), smem, stream,
-                       bfp(dz), bfp(w), bias, bfp_mut(y), (int)M, (int)N, (int)K);
-  } else if (M % 128 == 0 && K % 64 == 0 && getenv("GCBF_GEMM_NOGLDS") == nullptr) {
+    auto launch = [&](auto kernel) {
+      hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(dz), bfp(w), bias,
+                         bfp_mut(y), ya, (int)M, (int)N, (int)K);
+    };
+    if (actin == 1) launch(gemm_bias_act_sm_kernel<0, true, 1>);
+    else if (actin == 2) launch(gemm_bias_act_sm_kernel<0, true, 2>);
+    else launch(gemm_bias_act_sm_kernel<0, true, 0>);
+  } else if (actin == 0 && M % 128 == 0 && K % 64 == 0 &&
+             getenv("GCBF_GEMM_NOGLDS") == nullptr) {
     dim3 grid(M / 128, (N + 63) / 64);
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 2 * 128 * 64 * sizeof(uint16_t);
     hipLaunchKernelGGL((gemm_bias_act_glds_kernel<0, true>), grid, dim3(256), smem, stream,
@@ -221,8 +242,13 @@ torch::Tensor gemm_bt(torch::Tensor dz, torch::Tensor w) {
   } else {
     dim3 grid((M + 127) / 128, (N + 63) / 64);
     size_t smem = (size_t)(K / 8) * 64 * 8 * sizeof(uint16_t) + 128 * 40 * sizeof(uint16_t);
-    hipLaunchKernelGGL((gemm_bias_act_kernel<0, true>), grid, dim3(256), smem, stream,
-                       bfp(dz), bfp(w), bias, bfp_mut(y), (int)M, (int)N, (int)K);
+    auto launch = [&](auto kernel) {
+      hipLaunchKernelGGL(kernel, grid, dim3(256), smem, stream, bfp(dz), bfp(w), bias,
+                         bfp_mut(y), ya, (int)M, (int)N, (int)K);
+    };
+    if (actin == 1) launch(gemm_bias_act_kernel<0, true, 1>);
+    else if (actin == 2) launch(gemm_bias_act_kernel<0, true, 2>);
+    else launch(gemm_bias_act_kernel<0, true, 0>);
   }
   return y;
 }

@@ -63,11 +63,11 @@ __device__ __forceinline__ void stage_B_image(const bf16_t* __restrict__ W, bf16
 // LDS: B-block [K/8][64][8] (k-blocked so B-fragments are single b128 reads)
 //      + A-tile [128][40] (pad 32->40 kills b128 bank conflicts).
 // ---------------------------------------------------------------------------
-template <int ACT, bool BT>
+template <int ACT, bool BT, int ACTIN>
 __launch_bounds__(256) __global__
 void gemm_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
                           const float* __restrict__ bias, bf16_t* __restrict__ Y,
-                          int M, int N, int K) {
+                          const bf16_t* __restrict__ Xact, int M, int N, int K) {
   constexpr int BM = 128, BN = 64, BK = 32, APAD = 40;
   extern __shared__ char smem[];
   bf16_t* sB = (bf16_t*)smem;                    // [K/8][BN][8]
@@ -100,6 +100,15 @@ void gemm_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict
     if (arow_g < M) {
       *(bf16x8*)av = *(const bf16x8*)(X + arow_g * K + k0 + acol);
       *(bf16x8*)(av + 8) = *(const bf16x8*)(X + arow_g * K + k0 + acol + 8);
+      if constexpr (ACTIN != 0) {
+        // X operand is dY of an activated layer: fold dZ = dY * act'(Y)
+        bf16_t yv[16];
+        *(bf16x8*)yv = *(const bf16x8*)(Xact + arow_g * K + k0 + acol);
+        *(bf16x8*)(yv + 8) = *(const bf16x8*)(Xact + arow_g * K + k0 + acol + 8);
+#pragma unroll
+        for (int i = 0; i < 16; ++i)
+          av[i] = (bf16_t)((float)av[i] * act_grad_from_out((float)yv[i], ACTIN));
+      }
     } else {
 #pragma unroll
       for (int i = 0; i < 16; ++i) av[i] = (bf16_t)0.f;
@@ -240,11 +249,11 @@ template __global__ void gemm_bias_act_bn128_kernel<2>(const bf16_t*, const bf16
 // Small-M variant: BM=32 x BN=64 tile (4 waves as 2x2, wave 16x32) so
 // mid-size rows (update/head layers, M ~ 2k) still fill all 256 CUs.
 // ---------------------------------------------------------------------------
-template <int ACT, bool BT>
+template <int ACT, bool BT, int ACTIN>
 __launch_bounds__(256) __global__
 void gemm_bias_act_sm_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
                              const float* __restrict__ bias, bf16_t* __restrict__ Y,
-                             int M, int N, int K) {
+                             const bf16_t* __restrict__ Xact, int M, int N, int K) {
   constexpr int BM = 32, BN = 64, BK = 32, APAD = 40;
   extern __shared__ char smem[];
   bf16_t* sB = (bf16_t*)smem;                                        // [K/8][BN][8]
@@ -271,6 +280,13 @@ void gemm_bias_act_sm_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
     const long arow_g = (long)(m0 + arow);
     if (arow_g < M) {
       *(uint2*)av = *(const uint2*)(X + arow_g * K + k0 + acol);
+      if constexpr (ACTIN != 0) {
+        bf16_t yv[4];
+        *(uint2*)yv = *(const uint2*)(Xact + arow_g * K + k0 + acol);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          av[i] = (bf16_t)((float)av[i] * act_grad_from_out((float)yv[i], ACTIN));
+      }
     } else {
 #pragma unroll
       for (int i = 0; i < 4; ++i) av[i] = (bf16_t)0.f;
@@ -301,10 +317,12 @@ void gemm_bias_act_sm_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   }
 }
 
-template __global__ void gemm_bias_act_sm_kernel<0, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_sm_kernel<1, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_sm_kernel<2, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_sm_kernel<0, true>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<0, false, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<1, false, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<2, false, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<0, true, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<0, true, 1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_sm_kernel<0, true, 2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
 
 // ---------------------------------------------------------------------------
 // Small-N path (N <= 16): one wave per output row, W cached in LDS.
@@ -351,8 +369,10 @@ __global__ void act_bwd_kernel(const bf16_t* __restrict__ dY, const bf16_t* __re
 // A-operand = X^T staged transposed in LDS (scalar transpose writes),
 // B-operand = dZ staged k-blocked like the forward GEMM.
 // ---------------------------------------------------------------------------
+template <int ACT>
 __launch_bounds__(256) __global__
 void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
+                            const bf16_t* __restrict__ Yact,
                             float* __restrict__ partial, float* __restrict__ db_partial,
                             int M, int N, int K, int S) {
   // 64x64 output tile, BMR=64 reduction steps with register-prefetch
@@ -401,10 +421,22 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
       }
       if (mr < me && n0 + kc + 7 < N) {
         *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + mr * N + n0 + kc);
+        if constexpr (ACT != 0) {
+          // dZ operand is dY here: fold dZ = dY * act'(Y) into the stage
+          bf16x8 yv = *(const bf16x8*)(Yact + mr * N + n0 + kc);
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            zv[h][i] = (bf16_t)((float)zv[h][i] * act_grad_from_out((float)yv[i], ACT));
+        }
       } else {
 #pragma unroll
-        for (int i = 0; i < 8; ++i)
-          zv[h][i] = (mr < me && n0 + kc + i < N) ? dZ[mr * N + n0 + kc + i] : (bf16_t)0.f;
+        for (int i = 0; i < 8; ++i) {
+          const bool ok = mr < me && n0 + kc + i < N;
+          float z = ok ? (float)dZ[mr * N + n0 + kc + i] : 0.f;
+          if constexpr (ACT != 0)
+            if (ok) z *= act_grad_from_out((float)Yact[mr * N + n0 + kc + i], ACT);
+          zv[h][i] = (bf16_t)z;
+        }
       }
     }
   };
@@ -477,6 +509,10 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
     }
 }
 
+template __global__ void gemm_tn_partial_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+
 // (dw_partial (S,K,N), db_partial (S,N)) -> (dW, db) in ONE launch
 // (fixed-order sums: deterministic; 4 accumulators hide add latency).
 // acc != 0 accumulates (+=) into dW/db — used to write straight into the
@@ -507,10 +543,12 @@ __global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* _
 // ---------------------------------------------------------------------------
 // host-visible launchers (called from bindings.cpp)
 // ---------------------------------------------------------------------------
-template __global__ void gemm_bias_act_kernel<0, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_kernel<1, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_kernel<2, false>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
-template __global__ void gemm_bias_act_kernel<0, true>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<0, false, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<1, false, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<2, false, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<0, true, 0>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<0, true, 1>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
+template __global__ void gemm_bias_act_kernel<0, true, 2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
 template __global__ void gemv_bias_act_kernel<0, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
 template __global__ void gemv_bias_act_kernel<1, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);
 template __global__ void gemv_bias_act_kernel<2, bf16_t>(const bf16_t*, const bf16_t*, const float*, bf16_t*, int, int, int);

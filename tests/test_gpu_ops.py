@@ -84,8 +84,8 @@ def test_gemm_tn_deterministic():
     x = _rand(4096, 128).to(torch.bfloat16)
     dz = _rand(4096, 256).to(torch.bfloat16)
     from gcbfplus_amd import _C
-    dw1, db1 = _C.gemm_tn(x, dz)
-    dw2, db2 = _C.gemm_tn(x, dz)
+    dw1, db1 = _C.gemm_tn(x, dz, dz, 0)
+    dw2, db2 = _C.gemm_tn(x, dz, dz, 0)
     assert torch.equal(dw1, dw2) and torch.equal(db1, db2)
 
 

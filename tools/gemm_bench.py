@@ -26,11 +26,11 @@ def bench_tn(M, K, N, iters=50):
     x = torch.randn(M, K, device="cuda").to(torch.bfloat16)
     dz = torch.randn(M, N, device="cuda").to(torch.bfloat16)
     for _ in range(5):
-        dw, db = _C.gemm_tn(x, dz)
+        dw, db = _C.gemm_tn(x, dz, dz, 0)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        dw, db = _C.gemm_tn(x, dz)
+        dw, db = _C.gemm_tn(x, dz, dz, 0)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
     tf = 2 * M * K * N / dt / 1e12
