@@ -344,7 +344,14 @@ void gemv_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict
   float p[16];
 #pragma unroll
   for (int n = 0; n < 16; ++n) p[n] = 0.f;
-  for (int k = lane; k < K; k += WAVE) {
+  const int kvec = (K / 8) * 8;
+  for (int k8 = lane * 8; k8 < kvec; k8 += WAVE * 8) {  // b128 row loads
+    const bf16x8 xv = *(const bf16x8*)(X + m * K + k8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      for (int n = 0; n < N; ++n) p[n] += (float)xv[i] * (float)sW[(k8 + i) * N + n];
+  }
+  for (int k = kvec + lane; k < K; k += WAVE) {  // ragged tail
     const float xv = (float)X[m * K + k];
     for (int n = 0; n < N; ++n) p[n] += xv * (float)sW[k * N + n];
   }
