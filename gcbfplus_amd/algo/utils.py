@@ -71,7 +71,8 @@ def horizon_safe_mask(unsafe: Tensor, horizon: int) -> Tensor:
 
 def _dense_tree(d) -> Dict[str, np.ndarray]:
     return {
-        "kernel": d.kernel.detach().cpu().float().numpy(),
+        # export the logical rows only (Dense may pad its kernel to K%32==0)
+        "kernel": d.kernel.detach()[: d.in_dim].cpu().float().numpy(),
         "bias": d.bias.detach().cpu().float().numpy(),
     }
 
@@ -108,7 +109,12 @@ def net_to_flax_tree(net, head_name: str, out_name: str = "Dense_0") -> dict:
 
 def _load_dense(d, tree):
     with torch.no_grad():
-        d.kernel.copy_(torch.from_numpy(np.asarray(tree["kernel"], dtype=np.float32)))
+        src = torch.from_numpy(np.asarray(tree["kernel"], dtype=np.float32))
+        if src.shape[0] < d.kernel.shape[0]:  # padded-at-init kernel
+            d.kernel.zero_()
+            d.kernel[: src.shape[0]].copy_(src)
+        else:
+            d.kernel.copy_(src)
         d.bias.copy_(torch.from_numpy(np.asarray(tree["bias"], dtype=np.float32)))
 
 

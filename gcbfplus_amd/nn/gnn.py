@@ -50,7 +50,10 @@ class GNNLayer(nn.Module):
     def __init__(self, node_dim: int, edge_dim: int, msg_dim: int, out_dim: int,
                  hid_msg=(256, 256), hid_aggr=(128, 128), hid_update=(256, 256)):
         super().__init__()
-        self.msg_mlp = MLP(edge_dim + 2 * node_dim, hid_msg, act="relu", act_final=False)
+        in_msg = edge_dim + 2 * node_dim
+        pad = (in_msg + 31) // 32 * 32  # fused edge_msg_in emits this K
+        self.msg_mlp = MLP(in_msg, hid_msg, act="relu", act_final=False,
+                           pad_first_to=pad if pad != in_msg else None)
         self.msg_out = Dense(hid_msg[-1], msg_dim)
         self.attn_mlp = MLP(msg_dim, hid_aggr, act="relu", act_final=False)
         self.attn_out = Dense(hid_aggr[-1], 1)

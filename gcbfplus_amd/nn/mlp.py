@@ -20,16 +20,26 @@ _ACT = {"relu": ops.ACT_RELU, "tanh": ops.ACT_TANH, "none": ops.ACT_NONE}
 
 
 class Dense(nn.Module):
-    """y = act(x @ kernel + bias); kernel (in_dim, out_dim) fp32."""
+    """y = act(x @ kernel + bias); kernel (in_dim, out_dim) fp32.
 
-    def __init__(self, in_dim: int, out_dim: int, act: str = "none", scale: Optional[float] = None):
+    ``pad_to``: allocate the kernel with ``pad_to`` rows, zeroing rows beyond
+    ``in_dim``. The MFMA GEMM needs K % 32 == 0, and the fused edge-input op
+    emits K already padded — padding the PARAMETER once at init (instead of
+    cat-padding the weight every call) keeps the hot path allocation-free.
+    Zero rows stay zero under AdamW (grad 0, decay of 0); checkpoints
+    export/import the logical ``[:in_dim]`` slice (algo/utils.py)."""
+
+    def __init__(self, in_dim: int, out_dim: int, act: str = "none", scale: Optional[float] = None,
+                 pad_to: Optional[int] = None):
         super().__init__()
         self.in_dim, self.out_dim = in_dim, out_dim
         self.act = _ACT[act]
-        w = torch.empty(in_dim, out_dim)
+        rows = pad_to if pad_to is not None else in_dim
+        assert rows >= in_dim
+        w = torch.zeros(rows, out_dim)
         # xavier uniform on (out, in) fan convention == flax default_nn_init
         limit = math.sqrt(6.0 / (in_dim + out_dim))
-        nn.init.uniform_(w, -limit, limit)
+        nn.init.uniform_(w[:in_dim], -limit, limit)
         if scale is not None:
             w *= scale
         self.kernel = nn.Parameter(w)
@@ -54,6 +64,7 @@ class MLP(nn.Module):
         act: str = "relu",
         act_final: bool = True,
         scale_final: Optional[float] = None,
+        pad_first_to: Optional[int] = None,
     ):
         super().__init__()
         layers = []
@@ -62,7 +73,8 @@ class MLP(nn.Module):
         for i, h in enumerate(hid_sizes):
             last = i == n - 1
             layer_act = "none" if (last and not act_final) else act
-            layers.append(Dense(d, h, act=layer_act, scale=scale_final if last else None))
+            layers.append(Dense(d, h, act=layer_act, scale=scale_final if last else None,
+                                pad_to=pad_first_to if i == 0 else None))
             d = h
         self.layers = nn.ModuleList(layers)
         self.out_dim = d

@@ -164,11 +164,15 @@ def fused_linear(x: Tensor, w: Tensor, b: Optional[Tensor], act: int = ACT_NONE)
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
     if x2.shape[-1] != w.shape[0]:
-        # zero-padded input (fused edge_msg_in emits K padded to 32): pad W
-        # rows to match — zeros x zeros contribute nothing, and autograd
-        # slices dW back to the master shape
-        assert x2.shape[-1] > w.shape[0]
-        w = torch.cat([w, w.new_zeros(x2.shape[-1] - w.shape[0], w.shape[1])], dim=0)
+        if x2.shape[-1] < w.shape[0]:
+            # weight padded at init (Dense pad_to) but input unpadded (eager
+            # CPU edge path): the dropped rows are zero, so slicing is exact
+            w = w.narrow(0, 0, x2.shape[-1])
+        else:
+            # zero-padded input (fused edge_msg_in emits K padded to 32): pad
+            # W rows to match — zeros x zeros contribute nothing, and autograd
+            # slices dW back to the master shape
+            w = torch.cat([w, w.new_zeros(x2.shape[-1] - w.shape[0], w.shape[1])], dim=0)
     if x2.is_cuda:
         y = _FusedLinearHIP.apply(x2, w, b, act)
     else:
