@@ -20,8 +20,6 @@ __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
 template <int ACT>
 __global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
-template <int ACT>
-__global__ void gemm_tn_wide_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
@@ -167,10 +165,7 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   CHECK_IN(dz);
   long M = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == M);
-  // wide-K kernel (128xK block, 64x32 wave tile) when K allows; 64x64 else
-  const bool wide = K >= 128 && getenv("GCBF_TN_NOWIDE") == nullptr;
-  long gk = wide ? (K + 127) / 128 : (K + 63) / 64;
-  long gn = (N + 63) / 64;
+  long gk = (K + 63) / 64, gn = (N + 63) / 64;
   // deterministic split count: aim for ~1024 blocks, depends on shapes only
   long S = std::min<long>(128, std::max<long>(1, 1024 / std::max<long>(1, gk * gn)));
   S = std::min<long>(S, std::max<long>(1, (M + 31) / 32));
@@ -185,11 +180,7 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
                        bfp(x), bfp(dz), ya, partial.data_ptr<float>(),
                        db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
   };
-  if (wide) {
-    if (actin == 1) launch(gemm_tn_wide_kernel<1>);
-    else if (actin == 2) launch(gemm_tn_wide_kernel<2>);
-    else launch(gemm_tn_wide_kernel<0>);
-  } else if (actin == 1) launch(gemm_tn_partial_kernel<1>);
+  if (actin == 1) launch(gemm_tn_partial_kernel<1>);
   else if (actin == 2) launch(gemm_tn_partial_kernel<2>);
   else launch(gemm_tn_partial_kernel<0>);
   hipLaunchKernelGGL(reduce_dw_db_kernel, dim3((K * N + N + 255) / 256), dim3(256), 0, stream,

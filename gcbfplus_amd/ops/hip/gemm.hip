@@ -520,164 +520,6 @@ template __global__ void gemm_tn_partial_kernel<0>(const bf16_t*, const bf16_t*,
 template __global__ void gemm_tn_partial_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
 template __global__ void gemm_tn_partial_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
 
-
-// ---------------------------------------------------------------------------
-// Wide-K variant for K >= 128: 128(K) x 64(N) block, wave tile 64x32 so each
-// wave runs 16 MFMA per 12 LDS fragment loads (the 64x64 kernel's 32x32 wave
-// tile runs 8 per 8) — MFMA-throughput rather than LDS-load bound.
-// grid: (ceil(K/128), ceil(N/64), S); block 256 (4 waves 2x2).
-// ---------------------------------------------------------------------------
-template <int ACT>
-__launch_bounds__(256) __global__
-void gemm_tn_wide_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
-                         const bf16_t* __restrict__ Yact,
-                         float* __restrict__ partial, float* __restrict__ db_partial,
-                         int M, int N, int K, int S) {
-  constexpr int BKDIM = 128, BN = 64, BMR = 64, TPAD = 72;
-  __shared__ bf16_t sXT[BKDIM][TPAD];   // [k][m] transposed X tile
-  __shared__ bf16_t sB[BMR / 8][BN][8]; // dZ tile, m-blocked
-  __shared__ float sDb[4][BN];
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int w = tid >> 6;
-  const int wk = w >> 1, wn = w & 1;  // wave tile 64(K) x 32(N)
-  const int k0 = blockIdx.x * BKDIM;
-  const int n0 = blockIdx.y * BN;
-  const int s = blockIdx.z;
-
-  const long m_per = ((long)M + S - 1) / S;
-  const long ms = (long)s * m_per;
-  const long me = (ms + m_per < (long)M) ? ms + m_per : (long)M;
-
-  f32x4 acc[4][2];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
-  float db_acc = 0.f;
-  const int db_c = tid & 63, db_q = tid >> 6;
-
-  // staging: X 64 rows x 128 k-cols = 16 chunks/row -> 4 per thread;
-  //          dZ 64 rows x 64 n = 8 chunks/row -> 2 per thread
-  bf16_t xv[4][8], zv[2][8];
-
-  auto load_tile = [&](long m0) {
-#pragma unroll
-    for (int h = 0; h < 4; ++h) {
-      const int c = tid + h * 256;
-      const long mr = m0 + (c >> 4);
-      const int kc = (c & 15) * 8;
-      if (mr < me && k0 + kc + 7 < K) {
-        *(bf16x8*)xv[h] = *(const bf16x8*)(X + mr * K + k0 + kc);
-      } else {
-#pragma unroll
-        for (int i = 0; i < 8; ++i)
-          xv[h][i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
-      }
-    }
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      const int c = tid + h * 256;
-      const long mr = m0 + (c >> 3);
-      const int nc = (c & 7) * 8;
-      if (mr < me && n0 + nc + 7 < N) {
-        *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + mr * N + n0 + nc);
-        if constexpr (ACT != 0) {
-          bf16x8 yv = *(const bf16x8*)(Yact + mr * N + n0 + nc);
-#pragma unroll
-          for (int i = 0; i < 8; ++i)
-            zv[h][i] = (bf16_t)((float)zv[h][i] * act_grad_from_out((float)yv[i], ACT));
-        }
-      } else {
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          const bool ok = mr < me && n0 + nc + i < N;
-          float z = ok ? (float)dZ[mr * N + n0 + nc + i] : 0.f;
-          if constexpr (ACT != 0)
-            if (ok) z *= act_grad_from_out((float)Yact[mr * N + n0 + nc + i], ACT);
-          zv[h][i] = (bf16_t)z;
-        }
-      }
-    }
-  };
-
-  auto write_tile = [&]() {
-#pragma unroll
-    for (int h = 0; h < 4; ++h) {
-      const int c = tid + h * 256;
-      const int mr = c >> 4;
-      const int kc = (c & 15) * 8;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) sXT[kc + i][mr] = xv[h][i];
-    }
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      const int c = tid + h * 256;
-      const int mr = c >> 3;
-      const int nc = (c & 7) * 8;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) sB[mr >> 3][nc + i][mr & 7] = zv[h][i];
-    }
-  };
-
-  load_tile(ms);
-  for (long m0 = ms; m0 < me; m0 += BMR) {
-    __syncthreads();
-    write_tile();
-    __syncthreads();
-    if (m0 + BMR < me) load_tile(m0 + BMR);  // overlap with the MFMA below
-
-    if (blockIdx.x == 0) {
-#pragma unroll
-      for (int q = 0; q < 2; ++q) {
-        const bf16x8 v = *(const bf16x8*)(&sB[db_q * 2 + q][db_c][0]);
-#pragma unroll
-        for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
-      }
-    }
-#pragma unroll
-    for (int mm = 0; mm < 2; ++mm) {  // two 32-deep reduction steps
-      bf16x8 afr[4], bfr[2];
-#pragma unroll
-      for (int kf = 0; kf < 4; ++kf)
-        afr[kf] = *(const bf16x8*)(&sXT[wk * 64 + kf * 16 + (lane & 15)][mm * 32 + (lane >> 4) * 8]);
-#pragma unroll
-      for (int nf = 0; nf < 2; ++nf)
-        bfr[nf] = *(const bf16x8*)(&sB[mm * 4 + (lane >> 4)][wn * 32 + nf * 16 + (lane & 15)][0]);
-#pragma unroll
-      for (int kf = 0; kf < 4; ++kf)
-#pragma unroll
-        for (int nf = 0; nf < 2; ++nf)
-          acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
-    }
-  }
-
-  if (blockIdx.x == 0) {
-    sDb[db_q][db_c] = db_acc;
-    __syncthreads();
-    if (db_q == 0 && n0 + db_c < N)
-      db_partial[(long)s * N + n0 + db_c] =
-          (sDb[0][db_c] + sDb[1][db_c]) + (sDb[2][db_c] + sDb[3][db_c]);
-  }
-  float* out = partial + (long)s * K * N;
-#pragma unroll
-  for (int kf = 0; kf < 4; ++kf)
-#pragma unroll
-    for (int nf = 0; nf < 2; ++nf) {
-      const int col = n0 + wn * 32 + nf * 16 + (lane & 15);
-      if (col >= N) continue;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int krow = k0 + wk * 64 + kf * 16 + (lane >> 4) * 4 + r;
-        if (krow < K) out[(long)krow * N + col] = acc[kf][nf][r];
-      }
-    }
-}
-
-template __global__ void gemm_tn_wide_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_wide_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_wide_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-
 // (dw_partial (S,K,N), db_partial (S,N)) -> (dW, db) in ONE launch
 // (fixed-order sums: deterministic; 4 accumulators hide add latency).
 // acc != 0 accumulates (+=) into dW/db — used to write straight into the
