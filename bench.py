@@ -35,6 +35,9 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--num-agents", type=int, default=8)
     ap.add_argument("--cpu", action="store_true")
+    ap.add_argument("--torch-profile", type=str, default=None, metavar="TRACE.json",
+                    help="wrap the timed steps in torch.profiler and export a "
+                         "chrome trace (SURVEY.md §5.1 observability)")
     args = ap.parse_args()
 
     local_rank = dp.setup_from_env()
@@ -75,8 +78,21 @@ def main():
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for i in range(args.steps):
-        train_step(args.warmup + i)
+    if args.torch_profile:
+        from torch.profiler import ProfilerActivity, profile, record_function
+
+        acts = [ProfilerActivity.CPU] + ([ProfilerActivity.CUDA] if use_cuda else [])
+        with profile(activities=acts) as prof:
+            for i in range(args.steps):
+                with record_function(f"train_step_{i}"):
+                    train_step(args.warmup + i)
+            if use_cuda:
+                torch.cuda.synchronize()
+        if dp.rank() == 0:
+            prof.export_chrome_trace(args.torch_profile)
+    else:
+        for i in range(args.steps):
+            train_step(args.warmup + i)
     if use_cuda:
         torch.cuda.synchronize()
     if dp.is_active():

@@ -104,6 +104,12 @@ class GCBFPlus(GCBF):
                 if not last and self._graphed_mb().run(batch, mb_idx):
                     continue
                 info = self._update_minibatch(batch[mb_idx], want_info=last)
+        # grad norms of the reported (last) minibatch (reference logs them,
+        # gcbf_plus.py:439)
+        for name, opt in (("cbf", self.cbf_optim), ("actor", self.actor_optim)):
+            norm = getattr(opt, "last_norm", None)
+            if norm is not None:
+                info[f"grad_norm/{name}"] = float(norm)
         return info
 
     # ---- QP labels (reference :193-211, 299-352) ---------------------------
