@@ -77,3 +77,52 @@ def np2torch(x, device="cpu"):
     if isinstance(x, tuple) and hasattr(x, "_fields"):
         return type(x)(*[np2torch(v, device) for v in x])
     return x
+
+
+def chunk_vmap(fn, chunks: int):
+    """Apply a batched fn in ``chunks`` pieces to bound peak memory
+    (reference :96-114, used for huge eval batches). The torch build's fns
+    are natively batched, so this is just chunked apply + tree_merge."""
+
+    def wrapper(*args):
+        batch = None
+        for t in (args[0],) if isinstance(args[0], Tensor) else args[0]:
+            batch = t.shape[0] if isinstance(t, Tensor) else len(t)
+            break
+        idxs = np.array_split(np.arange(batch), chunks)
+        out = [fn(*[tree_index(a, torch.as_tensor(ix)) for a in args]) for ix in idxs]
+        return tree_merge(out)
+
+    return wrapper
+
+
+class MutablePatchCollection:
+    """matplotlib PatchCollection whose patches can be moved between frames
+    (reference :116-124). Defined lazily so headless installs without
+    matplotlib can still import this module."""
+
+    def __new__(cls, patches, *args, **kwargs):
+        import matplotlib.collections as mcollections
+
+        class _Mutable(mcollections.PatchCollection):
+            def __init__(self, patches_, *a, **kw):
+                self._paths = None
+                self.patches = patches_
+                mcollections.PatchCollection.__init__(self, patches_, *a, **kw)
+
+            def get_paths(self):
+                self.set_paths(self.patches)
+                return self._paths
+
+        return _Mutable(patches, *args, **kwargs)
+
+
+def save_anim(ani, path):
+    """Save a matplotlib animation with frame progress (reference :141-149)."""
+    total = getattr(ani, "_save_count", None) or 0
+
+    def cb(curr_frame: int, total_frames: int):
+        if total and curr_frame % max(1, total // 10) == 0:
+            print(f"  animating frame {curr_frame}/{total}", flush=True)
+
+    ani.save(str(path), progress_callback=cb)

@@ -55,3 +55,13 @@ def test_with_agent_states_differentiable():
     loss = g2.states.square().sum()
     loss.backward()
     assert a.grad is not None and torch.isfinite(a.grad).all()
+
+
+def test_chunk_vmap_equivalent():
+    """chunk_vmap (reference utils.py:96-114): chunked apply == full apply."""
+    import torch
+    from gcbfplus_amd.utils.utils import chunk_vmap
+
+    x = torch.randn(11, 3)
+    f = chunk_vmap(lambda t: t.square().sum(-1), 4)
+    assert torch.allclose(f(x), x.square().sum(-1))
