@@ -122,3 +122,77 @@ class MetricsLogger:
     def close(self):
         if self._f is not None:
             self._f.close()
+
+
+# ---- misc reference-parity helpers (reference trainer/utils.py:58-177) ----
+
+def has_any_nan(tree) -> bool:
+    """True if any tensor leaf contains a non-finite value (reference :58-63)."""
+    from ..utils.utils import tree_map
+
+    found = []
+    tree_map(lambda t: found.append(not torch.isfinite(t).all().item()), tree)
+    return any(found)
+
+
+def tree_copy(tree):
+    """Deep clone of a tensor tree (reference :78-79)."""
+    from ..utils.utils import tree_map
+
+    return tree_map(lambda t: t.detach().clone(), tree)
+
+
+def is_connected(host: str = "8.8.8.8", port: int = 53, timeout: float = 2.0) -> bool:
+    """Network reachability probe gating online logging (reference :100-109)."""
+    import socket
+
+    try:
+        socket.setdefaulttimeout(timeout)
+        socket.socket(socket.AF_INET, socket.SOCK_STREAM).connect((host, port))
+        return True
+    except OSError:
+        return False
+
+
+def centered_norm(vmin, vmax):
+    """Zero-centered colormap normalization (reference :171-177)."""
+    from matplotlib.colors import CenteredNorm
+
+    if isinstance(vmin, list):
+        vmin = min(vmin)
+    if isinstance(vmax, list):
+        vmax = max(vmax)
+    halfrange = max(abs(vmin), abs(vmax))
+    return CenteredNorm(0, halfrange)
+
+
+def plot_cbf(fig, cbf, env, graph, agent_id: int, x_dim: int = 0, y_dim: int = 1,
+             n_mesh: int = 30):
+    """Filled CBF contours over a 2D slice of one agent's state
+    (reference :112-146): moves agent ``agent_id`` over an x/y mesh, rebuilds
+    the graph per mesh point and evaluates h."""
+    from ..utils.graph import GraphBatch
+
+    b0 = 0
+    states = graph.states[b0]
+    mask = graph.mask[b0]
+    side = env.area_size
+    xs = torch.linspace(0, side, n_mesh)
+    ys = torch.linspace(0, side, n_mesh)
+    gx, gy = torch.meshgrid(xs, ys, indexing="xy")
+    grid_states = states[None, None].repeat(n_mesh, n_mesh, 1, 1)
+    grid_states[:, :, agent_id, x_dim] = gx.to(states.device)
+    grid_states[:, :, agent_id, y_dim] = gy.to(states.device)
+    flat = grid_states.reshape(n_mesh * n_mesh, *states.shape)
+    g = GraphBatch(states=flat,
+                   mask=mask[None].expand(n_mesh * n_mesh, *mask.shape).contiguous(),
+                   n_agents=env.num_agents, n_rays=env.n_rays)
+    with torch.no_grad():
+        h = cbf(g)[:, agent_id, 0].reshape(n_mesh, n_mesh).cpu().numpy()
+    ax = fig.gca()
+    x, y = np.meshgrid(xs.numpy(), ys.numpy())
+    cf = ax.contourf(x, y, h, levels=15, alpha=0.5, cmap="magma")
+    fig.colorbar(cf, ax=ax)
+    ax.contour(x, y, h, levels=[0.0], colors="blue")
+    ax.axis("off")
+    return fig

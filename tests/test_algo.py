@@ -198,3 +198,34 @@ def test_nn_utils_helpers():
     got = safe_get(arr, torch.tensor([0, 3, 4, -1]))
     assert torch.allclose(got[:2], arr[[0, 3]])
     assert torch.isnan(got[2]).all() and torch.isnan(got[3]).all()
+
+
+def test_trainer_parity_helpers():
+    """has_any_nan / tree_copy / centered_norm / plot_cbf (reference
+    trainer/utils.py:58-177)."""
+    import matplotlib
+
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+    import numpy as np
+
+    from gcbfplus_amd.algo.module.cbf import CBFNet
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.trainer.utils import (centered_norm, has_any_nan, plot_cbf,
+                                            tree_copy)
+
+    assert not has_any_nan({"a": torch.ones(3)})
+    assert has_any_nan([torch.tensor([1.0, float("nan")])])
+    t = {"x": torch.randn(4)}
+    c = tree_copy(t)
+    assert torch.equal(c["x"], t["x"]) and c["x"].data_ptr() != t["x"].data_ptr()
+    nrm = centered_norm([-1.0, -3.0], [2.0])
+    assert nrm.halfrange == 3.0 and nrm.vcenter == 0
+
+    env = make_env("DoubleIntegrator", num_agents=4, area_size=2.0, max_step=4)
+    g = env.reset(1, np.random.default_rng(0))
+    net = CBFNet(env.node_dim, env.edge_dim, 1)
+    cbf = lambda gr: net(gr, env.edge_feats(gr))
+    fig = plt.figure()
+    plot_cbf(fig, cbf, env, g, agent_id=0, n_mesh=6)
+    plt.close(fig)
