@@ -41,7 +41,8 @@ def test(args):
         gnn_layers = 1
 
     env = make_env(env_id, num_agents, area_size=area_size, max_step=args.max_step,
-                   num_obs=args.obs, n_rays=args.n_rays, device=device)
+                   num_obs=args.obs, n_rays=args.n_rays, max_travel=args.max_travel,
+                   device=device)
 
     if args.u_ref:
         act_fn = env.u_ref
@@ -50,7 +51,7 @@ def test(args):
         algo = make_algo(
             algo=algo_name, env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
             state_dim=env.state_dim, action_dim=env.action_dim, n_agents=num_agents,
-            gnn_layers=gnn_layers,
+            gnn_layers=gnn_layers, alpha=args.alpha,
         )
         if args.path is not None:
             model_path = os.path.join(args.path, "models")
@@ -125,7 +126,7 @@ def test(args):
             unsafe = env.collision_mask(g).reshape(ro.time_horizon, -1)
             out = render_video(ro, os.path.join(videos_dir, f"{stamp}_epi{epi}.gif"),
                                env, b=0, Ta_is_unsafe=unsafe, cbf_fn=cbf_fn,
-                               cbf_agent=args.cbf)
+                               cbf_agent=args.cbf, dpi=args.dpi)
             print("video:", out)
 
 
@@ -147,6 +148,9 @@ def main():
     parser.add_argument("--cpu", action="store_true", default=False)
     parser.add_argument("--log", type=str, default=None)
     parser.add_argument("--cbf", type=int, default=None)
+    parser.add_argument("--alpha", type=float, default=1.0)
+    parser.add_argument("--max-travel", type=float, default=None)
+    parser.add_argument("--dpi", type=int, default=100)
     parser.add_argument("--nojit-rollout", action="store_true", default=False,
                         help="streamed eval without storing rollouts (512+ agents)")
     parser.add_argument("--no-video", action="store_true", default=False)
