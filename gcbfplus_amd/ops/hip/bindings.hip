@@ -17,6 +17,8 @@ __global__ void gemm_bias_act_glds_kernel(const bf16_t_*, const bf16_t_*, const 
 __global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int, int);
 template <int ACT, typename OutT>
 __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, OutT*, int, int, int);
+template <int ACT>
+__global__ void dot_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, float*, long, int);
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
 template <int ACT>
 __global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
@@ -86,6 +88,17 @@ torch::Tensor gemm_bias_act(torch::Tensor x, torch::Tensor w, torch::Tensor bias
     // h_dot = (h_next - h)/dt finite difference where bf16 storage would
     // cancel catastrophically (bf16 ulp at |h|~1 is 0.008, signal ~0.03h)
     auto y = torch::empty({M, N}, x.options().dtype(torch::kFloat32));
+    if (N == 1) {  // thread-per-row dot: gate / CBF head
+      dim3 grid((M + 255) / 256);
+      auto launch = [&](auto kernel) {
+        hipLaunchKernelGGL(kernel, grid, dim3(256), 0, stream, bfp(x), bfp(w),
+                           bias.data_ptr<float>(), y.data_ptr<float>(), M, (int)K);
+      };
+      if (act == 0) launch(dot_bias_act_kernel<0>);
+      else if (act == 1) launch(dot_bias_act_kernel<1>);
+      else launch(dot_bias_act_kernel<2>);
+      return y;
+    }
     dim3 grid((M + 3) / 4);
     size_t smem = (size_t)K * N * sizeof(uint16_t);
     auto launch = [&](auto kernel) {

@@ -325,6 +325,39 @@ template __global__ void gemm_bias_act_sm_kernel<0, true, 1>(const bf16_t*, cons
 template __global__ void gemm_bias_act_sm_kernel<0, true, 2>(const bf16_t*, const bf16_t*, const float*, bf16_t*, const bf16_t*, int, int, int);
 
 // ---------------------------------------------------------------------------
+// N == 1 path (attention gate, CBF head): ONE THREAD per output row — the
+// wave-per-row gemv dispatches M/4 tiny workgroups and is dispatch-bound at
+// M ~ 3e5; here 256 rows share a workgroup and W (<= 384 values, uniform
+// across the wave) comes from the scalar cache.
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void dot_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ W,
+                         const float* __restrict__ bias, float* __restrict__ Y,
+                         long M, int K) {
+  const long m = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (m >= M) return;
+  float a0 = 0.f, a1 = 0.f;
+  const int kv = (K / 16) * 16;
+  int k = 0;
+  for (; k < kv; k += 16) {  // two b128 loads in flight per iteration
+    const bf16x8 x0 = *(const bf16x8*)(X + m * K + k);
+    const bf16x8 x1 = *(const bf16x8*)(X + m * K + k + 8);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      a0 += (float)x0[i] * (float)W[k + i];
+      a1 += (float)x1[i] * (float)W[k + 8 + i];
+    }
+  }
+  for (; k < K; ++k) a0 += (float)X[m * K + k] * (float)W[k];
+  Y[m] = apply_act(a0 + a1 + bias[0], ACT);
+}
+
+template __global__ void dot_bias_act_kernel<0>(const bf16_t*, const bf16_t*, const float*, float*, long, int);
+template __global__ void dot_bias_act_kernel<1>(const bf16_t*, const bf16_t*, const float*, float*, long, int);
+template __global__ void dot_bias_act_kernel<2>(const bf16_t*, const bf16_t*, const float*, float*, long, int);
+
+// ---------------------------------------------------------------------------
 // Small-N path (N <= 16): one wave per output row, W cached in LDS.
 // grid: ceil(M/4); block 256 (4 waves).
 // ---------------------------------------------------------------------------
