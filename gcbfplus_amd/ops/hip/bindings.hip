@@ -181,7 +181,9 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   long gk = (K + 63) / 64, gn = (N + 63) / 64;
   // deterministic split count: aim for ~1024 blocks, depends on shapes only
   long S = std::min<long>(128, std::max<long>(1, 1024 / std::max<long>(1, gk * gn)));
-  S = std::min<long>(S, std::max<long>(1, (M + 31) / 32));
+  // keep >= 128 rows per split so small-M dW calls don't pay 64x partial
+  // traffic (S still a pure function of shapes: deterministic)
+  S = std::min<long>(S, std::max<long>(1, (M + 127) / 128));
 
   auto opts = x.options().dtype(torch::kFloat32);
   auto partial = torch::empty({S, K, N}, opts);
