@@ -1,14 +1,24 @@
 // MFMA GEMM kernels for the GNN MLP stack.
 //
 // The networks are stacks of Dense layers with K, N in {16..384} and M up to
-// ~1e5 rows (batch * agents * edge-slots), bf16 inputs / f32 accumulate.
-// Three kernels:
-//   gemm_bias_act_kernel : Y = act(X @ W + b), 128x64 tile, BK=32,
-//                          B-operand staged once per block in a
-//                          fragment-blocked LDS image, A staged per K-tile.
-//   gemv_bias_act_kernel : N <= 16 head/gate outputs, wave-per-row.
-//   gemm_tn kernels      : dW = X^T @ dZ with a deterministic split-M
-//                          partial-sum pass + reduction (no atomics).
+// ~3e5 rows (batch * agents * edge-slots), bf16 inputs / f32 accumulate.
+// Kernel family (launcher picks by shape, bindings.hip):
+//   gemm_bias_act_kernel      : Y = act(X @ W + b), 128x64 tile, BK=32;
+//                               B staged once per block in a fragment-blocked
+//                               LDS image, A staged per K-tile. Template
+//                               <ACT, BT, ACTIN>: BT consumes W^T through a
+//                               transposed B-stage (backward dX without a
+//                               transpose copy); ACTIN folds the upstream
+//                               activation backward into the A-stage.
+//   gemm_bias_act_glds_kernel : BK=64 double-buffered direct-to-LDS A staging
+//                               (global_load_lds) for M%128==0, K%64==0.
+//   gemm_bias_act_sm_kernel   : 32x64 tile so mid-size M still fills 256 CUs.
+//   gemm_bias_act_bn128_kernel: BN=128 experiment (GCBF_GEMM_BN128 env).
+//   dot_bias_act_kernel       : N == 1 gate/head outputs, thread-per-row.
+//   gemv_bias_act_kernel      : 2 <= N <= 16 outputs, wave-per-row, f32 out.
+//   gemm_tn_partial + reduce  : dW = X^T @ dZ, deterministic split-M partials
+//                               + fixed-order reduction (no atomics), with
+//                               fused db and optional upstream-act fold.
 #include "common.h"
 
 
