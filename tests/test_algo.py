@@ -283,3 +283,22 @@ def test_trainer_loop_cpu(tmp_path):
 
     assert glob.glob(str(tmp_path / "**" / "*.jsonl"), recursive=True) or \
         glob.glob(str(tmp_path / "*.jsonl"))
+
+
+def test_two_layer_gnn_update_smoke():
+    """gnn_layers=2 trains end-to-end (general jacobian path + eager GNN
+    layer-1 input build)."""
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.trainer.utils import collect_rollout
+
+    torch.manual_seed(2)
+    env = make_env("DoubleIntegrator", num_agents=3, area_size=2.0, max_step=4)
+    algo = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim,
+                     n_agents=3, gnn_layers=2, batch_size=4, buffer_size=8,
+                     horizon=2, inner_epoch=1, seed=0)
+    rng = np.random.default_rng(8)
+    g = env.reset(1, rng)
+    ro = collect_rollout(env, algo.step, g)
+    info = algo.update(ro, 0)
+    assert all(np.isfinite(v) for v in info.values()), info
