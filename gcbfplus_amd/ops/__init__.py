@@ -10,10 +10,17 @@ raise instead of silently falling back to eager (per-project rule: the HIP
 path must be the one that runs on GPU).
 
 Op inventory (kernel numbering follows SURVEY.md §2.7):
-  fused_linear      K3/K4: MFMA GEMM + bias + activation, with autograd
+  fused_linear        K3/K4: MFMA GEMM + bias + activation, with autograd
+                      (backward: transposed-B dX, deterministic split-M dW,
+                      upstream-activation fold, direct grad accumulation)
   masked_softmax_aggr K3/K4: per-receiver masked softmax + weighted message sum
-  raytrace_rect     K1: 2D LiDAR fan vs rectangle set (no grad)
-  proxqp_solve      K11: batched dense QP (labels; no grad)
+  edge_msg_in         K2/K15: fused layer-0 GNN input build (+ analytic bwd)
+  raytrace_rect       K1: 2D LiDAR fan vs rectangle set (no grad)
+  gcbf_plus_loss      K10: all GCBF+ hinge losses + metrics in one kernel pair
+  di_loss_prep        K10 prologue: u_ref + action clamp + euler + [cur; next]
+  proxqp_solve        K11: batched dense QP (labels; no grad)
+  (env_step.hip K5-K8 is launched via env._step_fused; optimizer.hip K13 via
+  ops.optim.FusedAdamW)
 """
 from __future__ import annotations
 
