@@ -305,3 +305,28 @@ def test_crazyflie_update_smoke():
     ro = collect_rollout(env, algo.step, g)
     info = algo.update(ro, 0)
     assert np.isfinite(info["loss/total"])
+
+
+def test_render_video_2d_and_3d(tmp_path):
+    """env/plot.py renders tiny gifs for a 2D and a 3D env (reference
+    env/plot.py:24-109, 189-413)."""
+    import matplotlib
+
+    matplotlib.use("Agg")
+    from gcbfplus_amd.algo import make_algo  # noqa: F401 (env setup deps)
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.env.plot import render_video
+    from gcbfplus_amd.trainer.utils import collect_rollout
+
+    for env_id in ("DoubleIntegrator", "LinearDrone"):
+        env = make_env(env_id, num_agents=2, area_size=2.0, max_step=3)
+        rng = np.random.default_rng(4)
+        g0 = env.reset(1, rng)
+        ro = collect_rollout(env, env.u_ref, g0)
+        g = ro.graph_at(env)
+        unsafe = env.collision_mask(g).reshape(ro.time_horizon, -1)
+        out = render_video(ro, str(tmp_path / f"{env_id}.gif"), env, b=0,
+                           Ta_is_unsafe=unsafe, dpi=40)
+        import os
+
+        assert os.path.exists(out) and os.path.getsize(out) > 0
