@@ -64,10 +64,14 @@ class Trainer:
         assert params["save_interval"] > 0
         return params
 
-    def train(self):
+    def train(self, start_step: int = 0):
+        """Run the outer loop; ``start_step`` > 0 resumes a run restored via
+        ``algo.load_full`` (framework extra over the reference — SURVEY §5.4:
+        the reference checkpoints params only)."""
         start = time.time()
         pbar = tqdm(total=self.steps, ncols=80)
-        for step in range(self.steps + 1):
+        pbar.update(start_step)
+        for step in range(start_step, self.steps + 1):
             if step % self.eval_interval == 0:
                 eval_info = self.eval_step()
                 self.logger.log({**eval_info, "step": step}, step=self.update_steps)
@@ -80,6 +84,9 @@ class Trainer:
                 )
                 if self.save_log and step % self.save_interval == 0:
                     self.algo.save(self.model_dir, step)
+                    # full training state for --resume (optimizers, target
+                    # net, rng) — overwritten in place each save
+                    self.algo.save_full(os.path.join(self.model_dir, "resume.pt"), step)
 
             graph0 = self.env.reset(self.n_env_train, self.rng)
             rollout = collect_rollout(self.env, self.algo.step, graph0, self._ro_train)

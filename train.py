@@ -49,13 +49,18 @@ def train(args):
                       n_env_train=args.n_env_train, n_env_test=args.n_env_test,
                       seed=args.seed, params=train_params, save_log=not args.debug)
 
+    start_step = 0
+    if args.resume is not None:
+        start_step = algo.load_full(args.resume) + 1
+        print(f"> Resumed full training state from {args.resume} (step {start_step})")
+
     if not args.debug:
         os.makedirs(log_dir, exist_ok=True)
         with open(os.path.join(log_dir, "config.yaml"), "w") as f:
             yaml.dump(vars(args), f)
             yaml.dump(algo.config, f)
 
-    trainer.train()
+    trainer.train(start_step=start_step)
 
 
 def main():
@@ -87,6 +92,9 @@ def main():
     parser.add_argument("--eval-interval", type=int, default=1)
     parser.add_argument("--eval-epi", type=int, default=1)
     parser.add_argument("--save-interval", type=int, default=10)
+    parser.add_argument("--resume", type=str, default=None,
+                        help="path to a models/resume.pt from a previous run "
+                             "(full optimizer/target/rng state)")
     args = parser.parse_args()
     train(args)
 
