@@ -25,6 +25,9 @@ __global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf1
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
+__global__ void mb_gather_kernel(const float*, const bool*, const bool*, const bool*,
+                                 const float*, const long*, float*, bool*, bool*, bool*,
+                                 float*, int, int, int, int);
 __global__ void grad_norm_sq_partial_kernel(const float*, long, float*);
 __global__ void reduce_norm_kernel(const float*, int, float*);
 __global__ void adamw_flat_kernel(float*, const float*, float*, float*, bf16_t_*, const float*, const int*,
@@ -266,6 +269,28 @@ torch::Tensor gemm_bt(torch::Tensor dz, torch::Tensor w, torch::Tensor yact, lon
     else launch(gemm_bias_act_kernel<0, true, 0>);
   }
   return y;
+}
+
+void mb_gather(torch::Tensor states, torch::Tensor masks, torch::Tensor safe,
+               torch::Tensor unsafe, torch::Tensor u_qp, torch::Tensor idx,
+               torch::Tensor o_states, torch::Tensor o_masks, torch::Tensor o_safe,
+               torch::Tensor o_unsafe, torch::Tensor o_uqp) {
+  CHECK_IN(states);
+  CHECK_IN(idx);
+  long mb = idx.numel();
+  long n_state = states.size(1) * states.size(2);
+  long n_mask = masks.size(1) * masks.size(2);
+  long n_flag = safe.size(1);
+  long n_uqp = u_qp.size(1) * u_qp.size(2);
+  TORCH_CHECK(n_state % 4 == 0 && idx.dtype() == torch::kInt64);
+  hipLaunchKernelGGL(mb_gather_kernel, dim3(mb), dim3(256), 0, cur_stream(),
+                     states.data_ptr<float>(), masks.data_ptr<bool>(),
+                     safe.data_ptr<bool>(), unsafe.data_ptr<bool>(),
+                     u_qp.data_ptr<float>(), idx.data_ptr<long>(),
+                     o_states.data_ptr<float>(), o_masks.data_ptr<bool>(),
+                     o_safe.data_ptr<bool>(), o_unsafe.data_ptr<bool>(),
+                     o_uqp.data_ptr<float>(), (int)n_state, (int)n_mask,
+                     (int)n_flag, (int)n_uqp);
 }
 
 std::vector<torch::Tensor> softmax_aggr_fwd(torch::Tensor gate, torch::Tensor msg,
@@ -538,5 +563,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn_acc", &gemm_tn_acc, "gemm_tn accumulated (+=) into given f32 buffers");
   m.def("softmax_aggr_fwd", &softmax_aggr_fwd);
   m.def("softmax_aggr_bwd", &softmax_aggr_bwd);
+  m.def("mb_gather", &mb_gather, "fused 5-tensor minibatch gather (K18)");
   m.def("raytrace_rect", &raytrace_rect);
 }

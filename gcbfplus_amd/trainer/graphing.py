@@ -80,12 +80,22 @@ class GraphedMinibatchStep:
         if self.fb is None:
             self._alloc(batch)
         fb = self.fb
-        torch.index_select(batch.states, 0, idx, out=fb.states)
-        torch.index_select(batch.masks, 0, idx, out=fb.masks)
-        torch.index_select(batch.safe, 0, idx, out=fb.safe)
-        torch.index_select(batch.unsafe, 0, idx, out=fb.unsafe)
-        if fb.u_qp is not None:
-            torch.index_select(batch.u_qp, 0, idx, out=fb.u_qp)
+        from .. import ops
+
+        ext = ops._load_ext()
+        if (fb.u_qp is not None and ext is not None and hasattr(ext, "mb_gather")
+                and (batch.states.shape[-2] * batch.states.shape[-1]) % 4 == 0):
+            # K18: all five gathers in one kernel
+            ext.mb_gather(batch.states, batch.masks, batch.safe, batch.unsafe,
+                          batch.u_qp, idx, fb.states, fb.masks, fb.safe,
+                          fb.unsafe, fb.u_qp)
+        else:
+            torch.index_select(batch.states, 0, idx, out=fb.states)
+            torch.index_select(batch.masks, 0, idx, out=fb.masks)
+            torch.index_select(batch.safe, 0, idx, out=fb.safe)
+            torch.index_select(batch.unsafe, 0, idx, out=fb.unsafe)
+            if fb.u_qp is not None:
+                torch.index_select(batch.u_qp, 0, idx, out=fb.u_qp)
         from ..parallel import dp
 
         if self.graph is None:

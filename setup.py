@@ -22,6 +22,7 @@ setup(
                 "gcbfplus_amd/ops/hip/loss.hip",
                 "gcbfplus_amd/ops/hip/env_step.hip",
                 "gcbfplus_amd/ops/hip/loss_prep.hip",
+                "gcbfplus_amd/ops/hip/gather.hip",
                 "gcbfplus_amd/ops/hip/optimizer.hip",
                 "gcbfplus_amd/ops/hip/bindings.hip",
             ],

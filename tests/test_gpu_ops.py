@@ -578,3 +578,25 @@ def test_gpu_update_step_all_envs(env_name, n, area):
     ro = collect_rollout(env, algo.step, g)
     info = algo.update(ro, 0)
     assert all(np.isfinite(v) for v in info.values()), (env_name, info)
+
+
+def test_mb_gather_matches_index_select():
+    """K18 fused 5-tensor gather vs torch.index_select."""
+    from gcbfplus_amd import _C
+
+    torch.manual_seed(50)
+    Nb, V, S, N, D, nu, mb = 37, 272, 4, 8, 41, 2, 16
+    states = torch.randn(Nb, V, S, device="cuda")
+    masks = torch.rand(Nb, N, D, device="cuda") < 0.5
+    safe = torch.rand(Nb, N, device="cuda") < 0.5
+    unsafe = torch.rand(Nb, N, device="cuda") < 0.3
+    u_qp = torch.randn(Nb, N, nu, device="cuda")
+    idx = torch.randperm(Nb, device="cuda")[:mb]
+    outs = (torch.empty(mb, V, S, device="cuda"),
+            torch.empty(mb, N, D, dtype=torch.bool, device="cuda"),
+            torch.empty(mb, N, dtype=torch.bool, device="cuda"),
+            torch.empty(mb, N, dtype=torch.bool, device="cuda"),
+            torch.empty(mb, N, nu, device="cuda"))
+    _C.mb_gather(states, masks, safe, unsafe, u_qp, idx, *outs)
+    for out, src in zip(outs, (states, masks, safe, unsafe, u_qp)):
+        assert torch.equal(out, src[idx])
