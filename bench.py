@@ -24,8 +24,10 @@ from gcbfplus_amd.env import make_env
 from gcbfplus_amd.parallel import dp
 from gcbfplus_amd.trainer.utils import collect_rollout
 
-N_ENV_PER_GPU = 16
-T_HORIZON = 256
+# benchmark constants (BASELINE.json config) — env overrides exist ONLY for
+# fast CI smoke of the JSON contract, never for reported numbers
+N_ENV_PER_GPU = int(os.environ.get("GCBF_BENCH_ENVS", 16))
+T_HORIZON = int(os.environ.get("GCBF_BENCH_T", 256))
 
 
 def main():

@@ -302,3 +302,26 @@ def test_two_layer_gnn_update_smoke():
     ro = collect_rollout(env, algo.step, g)
     info = algo.update(ro, 0)
     assert all(np.isfinite(v) for v in info.values()), info
+
+
+def test_bench_contract_json(tmp_path):
+    """bench.py emits the driver-contract JSON line on CPU (tiny config)."""
+    import json
+    import subprocess
+    import sys
+
+    env_vars = dict(os.environ, GCBF_BENCH_ENVS="2", GCBF_BENCH_T="16")
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--cpu", "--steps", "1", "--warmup", "0",
+         "--num-agents", "2"],
+        capture_output=True, text=True, timeout=900, env=env_vars,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    )
+    line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in d, k
+    assert d["n_gpus"] == 1 and d["steps"] == 1 and d["scaling"] == "weak"
+    assert d["data"] == "synthetic" and d["higher_is_better"] is True
