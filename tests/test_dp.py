@@ -43,15 +43,15 @@ def _run_rank(rank, world, tmpdir, q, same_data):
         torch.distributed.destroy_process_group()
 
 
-def _spawn_and_collect(same_data, tmp_path):
+def _spawn_and_collect(same_data, tmp_path, world=2):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_run_rank, args=(r, 2, str(tmp_path), q, same_data))
-             for r in range(2)]
+    procs = [ctx.Process(target=_run_rank, args=(r, world, str(tmp_path), q, same_data))
+             for r in range(world)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(2):
+    for _ in range(world):
         rank, flat = q.get(timeout=300)
         results[rank] = flat
     for p in procs:
@@ -92,3 +92,11 @@ def test_dp2_equals_dp1_on_same_data(tmp_path):
     # identical data on both ranks -> mean grad == single-rank grad exactly?
     # averaging identical fp32 grads is exact, so DP=2 == DP=1 bitwise
     assert np.allclose(res[0], flat, atol=0), "DP=2 != DP=1 on identical data"
+
+
+@pytest.mark.timeout(500)
+def test_dp4_params_stay_in_sync(tmp_path):
+    """4-rank gloo sync (exercises >2-way all-reduce trees)."""
+    res = _spawn_and_collect(same_data=False, tmp_path=tmp_path, world=4)
+    for r in range(1, 4):
+        assert np.array_equal(res[0], res[r]), f"rank {r} diverged"
