@@ -89,3 +89,22 @@ def test_crazyflie_analytic_edge_jac_matches_general():
         algo.gnn_layers = 1
     assert torch.allclose(h1, h2, atol=1e-6)
     assert torch.allclose(J1, J2, atol=3e-5), (J1 - J2).abs().max()
+
+
+@pytest.mark.parametrize("env_id,n", [("SingleIntegrator", 4), ("LinearDrone", 3),
+                                      ("DubinsCar", 4)])
+def test_fast_jacobian_all_envs(env_id, n):
+    """Analytic edge→state jacobian chain vs full autograd for every env
+    family that advertises analytic_edge_jac."""
+    torch.manual_seed(3)
+    env = make_env(env_id, num_agents=n, area_size=2.0, max_step=4, device="cpu")
+    if not getattr(env, "analytic_edge_jac", True):
+        pytest.skip("general path env")
+    g = env.reset(2, np.random.default_rng(2))
+    algo = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim,
+                     n_agents=n, gnn_layers=1, batch_size=8, buffer_size=16,
+                     horizon=4, seed=0)
+    h, J_fast = algo.cbf_and_jacobian(g, algo.cbf_tgt)
+    J_ref = autograd_jacobian(env, lambda gr, e: algo.cbf_tgt(gr, e), g)
+    assert torch.allclose(J_fast, J_ref, atol=3e-5), (J_fast - J_ref).abs().max()
