@@ -23,7 +23,8 @@ from ..utils.graph import GraphBatch
 from .base import MultiAgentController
 from .module.cbf import CBFNet
 from .module.policy import DeterministicPolicyNet
-from .utils import (clip_grads_, net_from_flax_tree, net_to_flax_tree, step_if_finite)
+from .utils import (clip_grads_, load_flax_pickle, net_from_flax_tree,
+                    net_to_flax_tree, step_if_finite)
 
 
 class GCBF(MultiAgentController):
@@ -100,8 +101,17 @@ class GCBF(MultiAgentController):
         """Adam (reference gcbf.py:101,118); GPU: fused flat-buffer kernel
         with the clip+finite-guard folded in (K13)."""
         if self._use_fused_optim():
-            self.cbf_optim = FusedAdamW(self.cbf, self.lr_cbf, 0.0, self.max_grad_norm)
-            self.actor_optim = FusedAdamW(self.actor, self.lr_actor, 0.0, self.max_grad_norm)
+            # Both nets' grad buffers are slices of ONE flat tensor: DP then
+            # averages everything with a single all-reduce (no cat, no extra
+            # copies; the bucket is HIP-graph capturable).
+            dev = self._env.device
+            n_cbf = sum(p.numel() for p in self.cbf.parameters() if p.requires_grad)
+            n_act = sum(p.numel() for p in self.actor.parameters() if p.requires_grad)
+            self.dp_gbuf = torch.zeros(n_cbf + n_act, device=dev)
+            self.cbf_optim = FusedAdamW(self.cbf, self.lr_cbf, 0.0, self.max_grad_norm,
+                                        gflat_buf=self.dp_gbuf[:n_cbf])
+            self.actor_optim = FusedAdamW(self.actor, self.lr_actor, 0.0, self.max_grad_norm,
+                                          gflat_buf=self.dp_gbuf[n_cbf:])
         else:
             self.cbf_optim = torch.optim.Adam(self.cbf.parameters(), lr=self.lr_cbf)
             self.actor_optim = torch.optim.Adam(self.actor.parameters(), lr=self.lr_actor)
@@ -136,11 +146,14 @@ class GCBF(MultiAgentController):
             pickle.dump(net_to_flax_tree(self.cbf, "CBFHead", "Dense_0"), f)
 
     def load(self, load_dir: str, step: int):
+        """Loads either this framework's checkpoints or the reference's own
+        pretrained pickles (jax arrays unpickled jax-free via
+        load_flax_pickle)."""
         path = os.path.join(load_dir, str(step))
-        with open(os.path.join(path, "actor.pkl"), "rb") as f:
-            net_from_flax_tree(self.actor, pickle.load(f), "PolicyHead", "OutputDense")
-        with open(os.path.join(path, "cbf.pkl"), "rb") as f:
-            net_from_flax_tree(self.cbf, pickle.load(f), "CBFHead", "Dense_0")
+        net_from_flax_tree(self.actor, load_flax_pickle(os.path.join(path, "actor.pkl")),
+                           "PolicyHead", "OutputDense")
+        net_from_flax_tree(self.cbf, load_flax_pickle(os.path.join(path, "cbf.pkl")),
+                           "CBFHead", "Dense_0")
         self._refresh_optim_bf16()
 
     # ---- full training resume (NOT in the reference, which saves params
@@ -332,7 +345,7 @@ class GCBF(MultiAgentController):
         self.actor_optim.zero_grad(set_to_none=False)
         total.backward()
         if isinstance(self.cbf_optim, FusedAdamW):
-            dp.allreduce_mean_flat([self.cbf_optim.gflat, self.actor_optim.gflat])
+            dp.allreduce_mean_flat([self.dp_gbuf])
             cbf_norm = self.cbf_optim.step()
             actor_norm = self.actor_optim.step()
         else:

@@ -123,6 +123,32 @@ def _load_mlp(mlp, tree):
         _load_dense(l, tree[f"Dense_{i}"])
 
 
+def load_flax_pickle(path: str) -> dict:
+    """Unpickle a reference checkpoint (flax param tree pickled with jax
+    arrays, reference gcbf.py:344-357) WITHOUT jax installed: the only jax
+    global the pickles use is ``jax._src.array._reconstruct_array`` (verified
+    by opcode scan of /root/reference/pretrained/*/models/1000/*.pkl), whose
+    args carry a plain numpy reconstruction — map it to a numpy-returning
+    shim and load everything else normally."""
+    import pickle
+
+    def _reconstruct_np(fun, args, arr_state, aval_state):
+        arr = fun(*args)
+        arr.__setstate__(arr_state)
+        return np.asarray(arr)
+
+    class _U(pickle.Unpickler):
+        def find_class(self, module, name):
+            if module.startswith("jax") and name == "_reconstruct_array":
+                return _reconstruct_np
+            if module.startswith("jax"):
+                raise pickle.UnpicklingError(f"unexpected jax global {module}.{name}")
+            return super().find_class(module, name)
+
+    with open(path, "rb") as f:
+        return _U(f).load()
+
+
 def net_from_flax_tree(net, tree: dict, head_name: str, out_name: str = "Dense_0"):
     p = tree["params"]
     for i, layer in enumerate(net.gnn.layers):

@@ -23,13 +23,23 @@ from . import _require_ext
 
 class FusedAdamW:
     def __init__(self, module: nn.Module, lr: float, weight_decay: float = 0.0,
-                 max_grad_norm: float = float("inf"), betas=(0.9, 0.999), eps: float = 1e-8):
+                 max_grad_norm: float = float("inf"), betas=(0.9, 0.999), eps: float = 1e-8,
+                 gflat_buf: Optional[Tensor] = None):
         params = [p for p in module.parameters() if p.requires_grad]
         assert all(p.dtype == torch.float32 for p in params)
         device = params[0].device
         n = sum(p.numel() for p in params)
         self.pflat = torch.empty(n, device=device)
-        self.gflat = torch.zeros(n, device=device)
+        if gflat_buf is not None:
+            # caller-provided slice of a shared DP comm bucket: several
+            # optimizers' grads become ONE contiguous tensor so data-parallel
+            # needs exactly one all-reduce per minibatch (SURVEY §5.8:
+            # latency-bound tiny buckets — fewer calls beat overlap).
+            assert gflat_buf.numel() == n and gflat_buf.is_contiguous()
+            self.gflat = gflat_buf
+            self.gflat.zero_()
+        else:
+            self.gflat = torch.zeros(n, device=device)
         # bf16 shadow of pflat, updated in lock-step by the adamw kernel:
         # fused_linear reads p._bf instead of casting w per call (~28 cast
         # kernels per minibatch saved). Cold-path param mutations (loads,

@@ -7,6 +7,12 @@ fwd/bwd on its own shard; gradients are averaged with ONE fused-bucket
 all-reduce per minibatch (the two nets' grads are ~2.9 MB fp32 total, so
 latency dominates on 7x153 GB/s xGMI — a single flat bucket beats many small
 calls; overlap-with-backward is pointless at this size).
+
+RNG note (expected behavior, not a defect): each rank offsets its sampling
+stream (``seed + 1 + 7919*rank`` in ``algo/gcbf.py``) so DP ranks draw
+different replay minibatches and reset worlds. DP=1 matches the reference's
+single-device stream; DP>1 is not bit-reproducible against a single-device
+run by construction (the global batch differs — that is the point of DP).
 """
 from __future__ import annotations
 

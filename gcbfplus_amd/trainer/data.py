@@ -52,6 +52,24 @@ class Rollout(NamedTuple):
             env_states=obs,
         )
 
+    def graph_Tp1(self, env) -> GraphBatch:
+        """All T+1 graphs (pre-step states plus the terminal state) as one
+        GraphBatch of b*(T+1) — the reference's ``Tp1_graph`` eval semantics
+        (``/root/reference/test.py:184-186``, RolloutResult in env/base.py:25-31)."""
+        b, T = self.rewards.shape[:2]
+        obs = type(self.obstacles)(
+            *[f.repeat_interleave(T + 1, dim=0) for f in self.obstacles]
+        )
+        states = torch.cat([self.states, self.next_states[:, None]], dim=1)
+        masks = torch.cat([self.masks, self.next_mask[:, None]], dim=1)
+        return GraphBatch(
+            states=states.reshape(b * (T + 1), *states.shape[2:]),
+            mask=masks.reshape(b * (T + 1), *masks.shape[2:]),
+            n_agents=env.num_agents,
+            n_rays=env.n_rays,
+            env_states=obs,
+        )
+
 
 class FlatBatch(NamedTuple):
     """Flat per-timestep training samples (graphs only, obstacles not needed

@@ -57,8 +57,7 @@ class GraphedMinibatchStep:
 
     def _body(self):
         algo = self.algo
-        algo.cbf_optim.gflat.zero_()
-        algo.actor_optim.gflat.zero_()
+        algo.dp_gbuf.zero_()  # both optimizers' gflats are slices of this
         total, _ = algo._loss(self.fb, want_info=False)
         total.backward()
         if self.full:  # single-GPU: optimizer inside the capture too
@@ -112,8 +111,8 @@ class GraphedMinibatchStep:
                 opt.load_state_dict(sd)
         # stream capture records without executing -> always replay
         self.graph.replay()
-        if not self.full:  # DP: all-reduce between backward and the step
-            dp.allreduce_mean_flat([algo.cbf_optim.gflat, algo.actor_optim.gflat])
+        if not self.full:  # DP: ONE fused all-reduce between backward and the step
+            dp.allreduce_mean_flat([algo.dp_gbuf])
             algo.cbf_optim.step()
             algo.actor_optim.step()
         return True

@@ -98,6 +98,11 @@ class Trainer:
         self.logger.close()
 
     def eval_step(self) -> dict:
-        graph0 = self.env_test.reset(self.n_env_test, self.test_rng)
+        # Fixed eval worlds: the reference pre-splits one set of test keys
+        # before the loop (trainer/trainer.py:99-100) so every eval interval
+        # measures the same scenarios — re-seed per eval instead of advancing
+        # self.test_rng.
+        rng = np.random.default_rng(self.seed)
+        graph0 = self.env_test.reset(self.n_env_test, rng)
         rollout = collect_rollout(self.env_test, self.algo.act, graph0, self._ro_test)
         return eval_rollout_metrics(self.env_test, rollout)

@@ -17,6 +17,23 @@ from gcbfplus_amd.env import make_env
 from gcbfplus_amd.trainer.utils import collect_rollout
 
 
+def _load_run_config(f) -> dict:
+    """Load a run config.yaml: ours are plain mappings; the reference dumps an
+    argparse.Namespace with a python-object tag (loaded there via
+    yaml.UnsafeLoader, reference test.py:37-38). Support both without
+    arbitrary-object unpickling: only the Namespace tag is whitelisted."""
+
+    class _Loader(yaml.SafeLoader):
+        pass
+
+    def _ns(loader, node):
+        return loader.construct_mapping(node, deep=True)
+
+    _Loader.add_constructor(
+        "tag:yaml.org,2002:python/object:argparse.Namespace", _ns)
+    return yaml.load(f, Loader=_Loader)
+
+
 def test(args):
     print(f"> Running test.py {args}")
     np.random.seed(args.seed)
@@ -26,7 +43,7 @@ def test(args):
     # load config from the run dir (reference test.py:36-49)
     if args.path is not None:
         with open(os.path.join(args.path, "config.yaml")) as f:
-            config = yaml.safe_load(f)
+            config = _load_run_config(f)
         env_id = config.get("env") if args.env is None else args.env
         num_agents = config.get("num_agents") if args.num_agents is None else args.num_agents
         area_size = config.get("area_size") if args.area_size is None else args.area_size
@@ -85,8 +102,10 @@ def test(args):
             rollout = None
         else:
             rollout = collect_rollout(env, act_fn, graph0)
-            g = rollout.graph_at(env)
-            T = rollout.time_horizon
+            # Tp1 semantics: include the terminal (post-step) state, matching the
+            # reference test.py:184-186 and this file's streamed branch above.
+            g = rollout.graph_Tp1(env)
+            T = rollout.time_horizon + 1
             coll = env.collision_mask(g).reshape(1, T, -1)
             finish = env.finish_mask(g).reshape(1, T, -1)
             a_safe = 1.0 - coll.amax(dim=1).float()  # (1, N)

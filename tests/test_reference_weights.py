@@ -1,0 +1,74 @@
+"""Behavioral parity anchor: load the REFERENCE's own pretrained pickles
+(/root/reference/pretrained/<Env>/gcbf+/models/1000/{actor,cbf}.pkl, written
+by reference gcbf.py:344-349) into this framework's nets and check the
+policy actually works — this exercises the jax-free unpickler, the flax-tree
+layout mapping, and the full forward stack (edges, GNN, u_ref, dynamics,
+masks) end-to-end against weights trained in the reference implementation.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from gcbfplus_amd.algo import make_algo
+from gcbfplus_amd.algo.utils import load_flax_pickle
+from gcbfplus_amd.env import make_env
+from gcbfplus_amd.trainer.utils import collect_rollout
+
+REF = "/root/reference/pretrained"
+pytestmark = pytest.mark.skipif(not os.path.isdir(REF), reason="reference not present")
+
+
+def _algo_with_ref_weights(env_id, n_agents, area_size, device="cpu", **env_kw):
+    env = make_env(env_id, n_agents, area_size=area_size, device=device, **env_kw)
+    algo = make_algo(algo="gcbf+", env=env, node_dim=env.node_dim,
+                     edge_dim=env.edge_dim, state_dim=env.state_dim,
+                     action_dim=env.action_dim, n_agents=n_agents)
+    algo.load(os.path.join(REF, env_id, "gcbf+", "models"), 1000)
+    return env, algo
+
+
+def test_load_flax_pickle_no_jax():
+    tree = load_flax_pickle(os.path.join(REF, "DoubleIntegrator/gcbf+/models/1000/cbf.pkl"))
+    p = tree["params"]
+    assert set(p) == {"GNN_0", "CBFHead", "Dense_0"}
+    k = p["GNN_0"]["GNNLayer_0"]["msg"]["Dense_0"]["kernel"]
+    assert isinstance(k, np.ndarray) and k.dtype == np.float32 and k.ndim == 2
+
+
+def test_reference_di_weights_drive_safely():
+    """Reference DI checkpoint in this framework: zero collisions and most
+    agents reach goals within one episode (reference claims high safe/reach
+    at n=8; measured here 100%/94% over 4 epi — see BASELINE.md)."""
+    env, algo = _algo_with_ref_weights("DoubleIntegrator", 8, 4.0)
+    rng = np.random.default_rng(1234)
+    g = env.reset(1, rng)
+    ro = collect_rollout(env, algo.act, g)
+    gall = ro.graph_Tp1(env)
+    T = ro.time_horizon + 1
+    coll = env.collision_mask(gall).reshape(1, T, -1)
+    fin = env.finish_mask(gall).reshape(1, T, -1)
+    safe = 1.0 - coll.amax(dim=1).float()
+    finish = fin.amax(dim=1).float()
+    assert safe.mean().item() == 1.0, "reference weights should be collision-free"
+    assert finish.mean().item() >= 0.5, "most agents should reach their goals"
+
+
+def test_reference_weights_h_sign_sanity():
+    """h from the reference CBF weights: positive at a spread-out reset,
+    negative when two agents are forced into collision."""
+    env, algo = _algo_with_ref_weights("DoubleIntegrator", 8, 4.0)
+    rng = np.random.default_rng(7)
+    g = env.reset(1, rng)
+    h0 = algo.get_cbf(g)
+    assert torch.isfinite(h0).all()
+    assert (h0 > 0).float().mean().item() > 0.9
+    # drive agents 0 and 1 to the same point -> h must flag danger
+    states = g.states.clone()
+    states[0, 1, :2] = states[0, 0, :2] + 0.01
+    g2 = env.get_graph(states[:, : env.num_agents, :],
+                       states[:, env.num_agents : 2 * env.num_agents, :],
+                       g.env_states)
+    h2 = algo.get_cbf(g2)
+    assert (h2[0, :2] < 0).any(), "colliding pair should get negative h"
