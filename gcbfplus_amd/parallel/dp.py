@@ -44,7 +44,9 @@ def setup_from_env() -> int:
         dist.init_process_group(backend=backend)
     local = int(os.environ.get("LOCAL_RANK", 0))
     if torch.cuda.is_available():
-        torch.cuda.set_device(local)
+        # modulo so an over-subscribed validation run (2 ranks sharing the
+        # one leased GPU) still maps to a real device
+        torch.cuda.set_device(local % torch.cuda.device_count())
     return local
 
 

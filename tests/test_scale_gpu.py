@@ -1,0 +1,122 @@
+"""Scale-up GPU correctness: the shapes the real benchmark/large-swarm runs
+hit (n=64/128 agents, D≈1.5k edge slots, training-size GEMM M) — VERDICT
+round-1 weak #5: the small-shape suite never exercised these regimes, where
+softmax_aggr's one-WG-per-row layout and the edge kernels' occupancy
+assumptions could break.
+"""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from gcbfplus_amd import ops
+
+
+def _rand(*shape, scale=1.0):
+    return (torch.randn(*shape) * scale).to("cuda")
+
+
+@pytest.mark.parametrize("n_agents", [64, 128])
+def test_large_agents_cbf_forward_gpu_vs_cpu(n_agents):
+    """Full CBF forward at large N: GPU HIP path vs CPU fp32 compose."""
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.algo.module.cbf import CBFNet
+
+    torch.manual_seed(60)
+    env_c = make_env("DoubleIntegrator", num_agents=n_agents,
+                     area_size=float(max(4, n_agents // 8)), max_step=4,
+                     device="cpu")
+    g = env_c.reset(2, np.random.default_rng(61))
+    net = CBFNet(env_c.node_dim, env_c.edge_dim, 1)
+    with torch.no_grad():
+        h_cpu = net(g, env_c.edge_feats(g))
+        h_gpu = net.to("cuda")(g.to("cuda"), env_c.edge_feats(g.to("cuda")))
+    err = (h_gpu.cpu() - h_cpu).abs()
+    assert err.max() < 0.05, (n_agents, err.max())
+
+
+def test_softmax_aggr_large_D():
+    """D ≈ 1.5k slots (the 512-agent swarm regime claimed in
+    softmax_aggr.hip's header) — fwd + bwd vs the CPU fp32 path."""
+    torch.manual_seed(62)
+    B, N, D, C = 2, 4, 1537, 128
+    gate = _rand(B, N, D).requires_grad_(True)
+    msg = _rand(B, N, D, C, scale=0.5).requires_grad_(True)
+    mask = torch.rand(B, N, D, device="cuda") < 0.7
+    mask[:, :, 0] = True
+    out = ops.masked_softmax_aggr(gate, msg, mask)
+    gup = torch.randn_like(out)
+    out.backward(gup)
+
+    gate_c = gate.detach().cpu().requires_grad_(True)
+    msg_c = msg.detach().cpu().requires_grad_(True)
+    ref = ops.masked_softmax_aggr(gate_c, msg_c, mask.cpu())
+    ref.backward(gup.float().cpu())
+    err_f = (out.float().cpu() - ref).abs()
+    assert err_f.mean() < 5e-3 and err_f.max() < 5e-2, (err_f.mean(), err_f.max())
+    for got, refg in ((gate.grad, gate_c.grad), (msg.grad, msg_c.grad)):
+        err = (got.float().cpu() - refg).abs()
+        denom = refg.abs().mean().clamp_min(1e-4)
+        assert (err.mean() / denom) < 3e-2, (err.mean(), err.max())
+
+
+def test_training_shape_gemms():
+    """fused_linear at the REAL minibatch shapes of the benchmark run
+    (M = batch*N*D = 256*8*82 edge rows) vs fp32 torch matmul on-device."""
+    torch.manual_seed(63)
+    for M, K, N, act in [(167936, 10, 256, ops.ACT_RELU),
+                         (167936, 256, 128, 0),
+                         (2304, 384, 256, ops.ACT_RELU)]:
+        x = _rand(M, K, scale=0.5)
+        w = _rand(K, N, scale=0.2)
+        b = _rand(N, scale=0.1).float()
+        y = ops.fused_linear(x, w, b, act)
+        y_ref = ops._apply_act(x.float() @ w.float() + b, act)
+        err = (y.float() - y_ref).abs()
+        scale = y_ref.abs().mean().clamp_min(1.0)
+        assert (err.mean() / scale) < 5e-3, (M, K, N, err.mean().item())
+        assert (err.max() / scale) < 8e-2, (M, K, N, err.max().item())
+        del x, w, b, y, y_ref
+
+
+def test_gpu_update_step_n64():
+    """One full GCBF+ update at n=64 agents on GPU — all kernels at
+    large-graph occupancy."""
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.algo import make_algo
+    from gcbfplus_amd.trainer.utils import collect_rollout
+
+    torch.manual_seed(64)
+    env = make_env("DoubleIntegrator", num_agents=64, area_size=8.0, max_step=8,
+                   device="cuda")
+    algo = make_algo("gcbf+", env=env, node_dim=env.node_dim, edge_dim=env.edge_dim,
+                     state_dim=env.state_dim, action_dim=env.action_dim,
+                     n_agents=64, gnn_layers=1, batch_size=8, buffer_size=16,
+                     horizon=4, inner_epoch=1, seed=0)
+    rng = np.random.default_rng(65)
+    g = env.reset(2, rng)
+    ro = collect_rollout(env, algo.step, g)
+    info = algo.update(ro, 0)
+    assert all(np.isfinite(v) for v in info.values()), info
+
+
+def test_streamed_rollout_n256_di():
+    """Large-swarm streamed eval path (no stored graphs) at n=256 on GPU:
+    finite, and peak HBM stays far under the 288 GB budget."""
+    from gcbfplus_amd.env import make_env
+
+    torch.cuda.reset_peak_memory_stats()
+    env = make_env("DoubleIntegrator", num_agents=256, area_size=16.0,
+                   max_step=16, device="cuda")
+    rng = np.random.default_rng(66)
+    g = env.reset(1, rng)
+    ever_coll = env.collision_mask(g).float()
+    with torch.no_grad():
+        for _ in range(16):
+            a = env.u_ref(g)
+            g = env.step(g, a).graph
+            ever_coll = torch.maximum(ever_coll, env.collision_mask(g).float())
+    assert torch.isfinite(g.states).all()
+    peak_gb = torch.cuda.max_memory_allocated() / 2**30
+    assert peak_gb < 64, f"streamed n=256 rollout used {peak_gb:.1f} GB"
