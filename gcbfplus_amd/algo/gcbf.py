@@ -101,20 +101,23 @@ class GCBF(MultiAgentController):
         """Adam (reference gcbf.py:101,118); GPU: fused flat-buffer kernel
         with the clip+finite-guard folded in (K13)."""
         if self._use_fused_optim():
-            # Both nets' grad buffers are slices of ONE flat tensor: DP then
-            # averages everything with a single all-reduce (no cat, no extra
-            # copies; the bucket is HIP-graph capturable).
-            dev = self._env.device
-            n_cbf = sum(p.numel() for p in self.cbf.parameters() if p.requires_grad)
-            n_act = sum(p.numel() for p in self.actor.parameters() if p.requires_grad)
-            self.dp_gbuf = torch.zeros(n_cbf + n_act, device=dev)
-            self.cbf_optim = FusedAdamW(self.cbf, self.lr_cbf, 0.0, self.max_grad_norm,
-                                        gflat_buf=self.dp_gbuf[:n_cbf])
-            self.actor_optim = FusedAdamW(self.actor, self.lr_actor, 0.0, self.max_grad_norm,
-                                          gflat_buf=self.dp_gbuf[n_cbf:])
+            self._make_fused_optimizers(weight_decay=0.0)
         else:
             self.cbf_optim = torch.optim.Adam(self.cbf.parameters(), lr=self.lr_cbf)
             self.actor_optim = torch.optim.Adam(self.actor.parameters(), lr=self.lr_actor)
+
+    def _make_fused_optimizers(self, weight_decay: float):
+        """Both nets' grad buffers are slices of ONE flat tensor (dp_gbuf):
+        DP then averages everything with a single all-reduce (no cat, no
+        extra copies; the bucket is HIP-graph capturable)."""
+        dev = self._env.device
+        n_cbf = sum(p.numel() for p in self.cbf.parameters() if p.requires_grad)
+        n_act = sum(p.numel() for p in self.actor.parameters() if p.requires_grad)
+        self.dp_gbuf = torch.zeros(n_cbf + n_act, device=dev)
+        self.cbf_optim = FusedAdamW(self.cbf, self.lr_cbf, weight_decay,
+                                    self.max_grad_norm, gflat_buf=self.dp_gbuf[:n_cbf])
+        self.actor_optim = FusedAdamW(self.actor, self.lr_actor, weight_decay,
+                                      self.max_grad_norm, gflat_buf=self.dp_gbuf[n_cbf:])
 
     # ---- config / io -----------------------------------------------------
     @property

@@ -43,11 +43,8 @@ class GCBFPlus(GCBF):
 
     def _create_optimizers(self):
         """AdamW wd=1e-3 (reference gcbf_plus.py:109,127)."""
-        from ..ops.optim import FusedAdamW
-
         if self._use_fused_optim():
-            self.cbf_optim = FusedAdamW(self.cbf, self.lr_cbf, 1e-3, self.max_grad_norm)
-            self.actor_optim = FusedAdamW(self.actor, self.lr_actor, 1e-3, self.max_grad_norm)
+            self._make_fused_optimizers(weight_decay=1e-3)
         else:
             self.cbf_optim = torch.optim.AdamW(self.cbf.parameters(), lr=self.lr_cbf,
                                                weight_decay=1e-3)
@@ -171,8 +168,12 @@ class GCBFPlus(GCBF):
         analytically (reference does N reverse passes via jax.jacobian,
         gcbf_plus.py:310-317).
 
-        General path (gnn_layers > 1): N backward passes (row i of the
-        jacobian from grad of sum_m h[m, i]).
+        General path (gnn_layers > 1): ONE backward over an N-replicated
+        batch — replica (m, i) selects output h_i, so the whole (N, N)
+        jacobian block falls out of a single fused fwd+bwd instead of N
+        sequential reverse passes (the reference pays N passes via
+        jax.jacobian, gcbf_plus.py:310-317; callers chunk M, so the N-fold
+        activation memory is bounded).
         """
         env = self._env
         M, N, S = graph.batch_size, self.n_agents, graph.state_dim
@@ -184,16 +185,19 @@ class GCBFPlus(GCBF):
                 (ge,) = torch.autograd.grad(h.sum(), e)  # (M, N, D, E)
                 h_x = env.edge_grad_to_state_jac(graph, states, ge)
                 return h.detach(), h_x
-            # general fallback: one backward per receiver row
-            st = states.clone().requires_grad_(True)
-            e = env.edge_feats(graph, st)
-            h = cbf_net(graph, e).squeeze(-1)
-            rows = []
-            for i in range(N):
-                (gs,) = torch.autograd.grad(h[:, i].sum(), st, retain_graph=i < N - 1)
-                rows.append(gs[:, :N])  # only agent-state grads
-            h_x = torch.stack(rows, dim=1)  # (M, N, N, S)
-        return h.detach(), h_x
+            # batched-replica path: (M*N, V, S), replica k = m*N + i
+            st = states.repeat_interleave(N, dim=0).requires_grad_(True)
+            g_rep = GraphBatch(states=st,
+                               mask=graph.mask.repeat_interleave(N, dim=0),
+                               n_agents=N, n_rays=graph.n_rays)
+            e = env.edge_feats(g_rep, st)
+            h_rep = cbf_net(g_rep, e).squeeze(-1)  # (M*N, N)
+            row = torch.arange(N, device=st.device).repeat(M)
+            sel = h_rep.gather(1, row[:, None]).sum()
+            (gs,) = torch.autograd.grad(sel, st)
+            h_x = gs[:, :N].reshape(M, N, N, S)
+            h = h_rep.detach().reshape(M, N, N)[:, 0]
+        return h, h_x
 
     # ---- loss (reference :354-431) ----------------------------------------
     def _loss(self, mb: FlatBatch, want_info: bool = True) -> Tuple[Tensor, dict]:

@@ -448,6 +448,12 @@ class CrazyFlie(LinearDrone):
     # lidar: 3D fan over params['n_rays'] beams, top-n_hit returns
     def get_lidar_hits(self, agent_pos: Tensor, obstacles: Sphere) -> Tensor:
         from .utils import get_lidar
+        from .. import ops
 
+        if agent_pos.is_cuda and ops.hip_available() and obstacles.n_obs > 0:
+            return ops.raytrace_sphere_topk(
+                agent_pos, obstacles.center, obstacles.radius,
+                self._params["n_rays"], self._n_hit,
+                self._params["comm_radius"])
         return get_lidar(agent_pos, obstacles, self._params["n_rays"],
                          self._params["comm_radius"], max_returns=self._n_hit)

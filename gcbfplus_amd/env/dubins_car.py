@@ -110,12 +110,34 @@ class DubinsCar(DoubleIntegrator):
         return self.clip_state(graph.agent_states + x_dot * self._dt)
 
     def step(self, graph: GraphBatch, action: Tensor) -> StepResult:
+        if not __import__("os").environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is DubinsCar and self.enable_stop:
+            return self._step_fused(graph, action)
         next_agent = self._step_states(graph, action)
         reward = -((self.clip_action(action) - self.u_ref(graph)).square().sum(-1)).mean(-1)
         cost = self.get_cost(graph)
         done = torch.zeros(graph.batch_size, dtype=torch.bool, device=graph.device)
         next_graph = self.get_graph(next_agent, graph.goal_states, graph.env_states)
         return StepResult(next_graph, reward, cost, done, {})
+
+    def _step_fused(self, graph: GraphBatch, action: Tensor) -> StepResult:
+        """One-kernel env step on GPU (K5-K8, env_step2d_kernel<1>):
+        PID u_ref + stop freeze + euler + LiDAR rescan + masks."""
+        from .. import ops
+
+        ext = ops._require_ext()
+        p = self._params
+        if not hasattr(self, "_K_dummy"):
+            self._K_dummy = torch.zeros(2, 4, device=graph.device)
+        nxt, mask, reward, cost = ext.di_env_step(
+            graph.states.contiguous(), action.contiguous(),
+            graph.env_states.points.contiguous(), self._K_dummy,
+            self.num_agents, self.n_rays, self._dt, 0.0, p["comm_radius"],
+            p["car_radius"], 0.8, 1,
+        )
+        done = torch.zeros(graph.batch_size, dtype=torch.bool, device=graph.device)
+        g = GraphBatch(states=nxt, mask=mask, n_agents=self.num_agents,
+                       n_rays=self.n_rays, env_states=graph.env_states)
+        return StepResult(g, reward, cost, done, {})
 
     def forward_graph(self, graph: GraphBatch, action: Tensor) -> GraphBatch:
         return graph.with_agent_states(self._step_states(graph, action))

@@ -16,6 +16,7 @@ import numpy as np
 import torch
 from torch import Tensor
 
+from .. import ops
 from ..utils.graph import GraphBatch
 from .double_integrator import DoubleIntegrator
 from .obstacle import Sphere
@@ -125,6 +126,11 @@ class LinearDrone(DoubleIntegrator):
 
     # ---- lidar: 3D fan with top-16 selection (reference :290-299) -----------
     def get_lidar_hits(self, agent_pos: Tensor, obstacles: Sphere) -> Tensor:
+        if agent_pos.is_cuda and ops.hip_available() and obstacles.n_obs > 0:
+            return ops.raytrace_sphere_topk(
+                agent_pos, obstacles.center, obstacles.radius,
+                self._params["n_rays"], self.N_HIT_RETURNS,
+                self._params["comm_radius"])
         return get_lidar(agent_pos, obstacles, self._params["n_rays"],
                          self._params["comm_radius"], max_returns=self.N_HIT_RETURNS)
 
@@ -136,6 +142,40 @@ class LinearDrone(DoubleIntegrator):
         collision = (dist < 2 * r).any(dim=-1).float().mean(-1)
         inside = graph.env_states.inside(pos, r=r).float().mean(-1)
         return collision + inside
+
+    # ---- fused GPU step: part A (drone3d_step kernel: u_ref/reward/euler/
+    # cost/aa+goal mask) + part B (raytrace_sphere_topk in graph mode: hit
+    # node rows + lidar mask), replacing ~40 eager launches per step --------
+    def step(self, graph: GraphBatch, action: Tensor):
+        if not __import__("os").environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is LinearDrone and ops.hip_available() \
+                and graph.env_states.n_obs > 0:
+            return self._step_fused(graph, action)
+        return super().step(graph, action)
+
+    def _step_fused(self, graph: GraphBatch, action: Tensor):
+        from .base import StepResult
+
+        ext = ops._require_ext()
+        p = self._params
+        obs = graph.env_states
+        nxt, mask, reward, cost = ext.drone3d_step(
+            graph.states.contiguous(), action.contiguous(),
+            obs.center.contiguous(), obs.radius.contiguous(),
+            self._K.to(graph.device).contiguous(),
+            self._A_t.to(graph.device).contiguous(),
+            self.num_agents, self.n_rays, self._dt, 10.0, p["comm_radius"],
+            p["drone_radius"], 0.5,
+        )
+        # part B: lidar rows + lidar mask from the NEXT agent positions
+        next_pos = nxt[:, : self.num_agents, :3].contiguous()
+        ext.raytrace_sphere_graph(next_pos, obs.center.contiguous(),
+                                  obs.radius.contiguous(), nxt, mask,
+                                  p["n_rays"], self.N_HIT_RETURNS,
+                                  p["comm_radius"])
+        done = torch.zeros(graph.batch_size, dtype=torch.bool, device=graph.device)
+        g = GraphBatch(states=nxt, mask=mask, n_agents=self.num_agents,
+                       n_rays=self.n_rays, env_states=obs)
+        return StepResult(g, reward, cost, done, {})
 
     # ---- masks (reference :346-404): velocity-free margins ------------------
     def safe_mask(self, graph: GraphBatch) -> Tensor:

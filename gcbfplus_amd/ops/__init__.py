@@ -320,6 +320,19 @@ def raytrace_rect(pos: Tensor, points: Tensor, n_rays: int, sense_range: float) 
     raise NotImplementedError("CPU path goes through env.get_lidar composition")
 
 
+def raytrace_sphere_topk(pos: Tensor, centers: Tensor, radii: Tensor, n_beams: int,
+                         topk: int, sense_range: float) -> Tensor:
+    """3D LiDAR theta x phi fan + poles vs spheres, fused with stable top-k
+    closest-hit selection (reference env/utils.py:49-79 + obstacle.py:237-270
+    + the argsort top-k of utils.py:127-131). pos (B,N,3), centers (B,K,3),
+    radii (B,K) -> hits (B,N,topk,3). GPU only (no grad)."""
+    if pos.is_cuda:
+        _require_ext()
+        return _EXT.raytrace_sphere_topk(pos.contiguous(), centers.contiguous(),
+                                         radii.contiguous(), n_beams, topk, sense_range)
+    raise NotImplementedError("CPU path goes through env.get_lidar composition")
+
+
 # --------------------------------------------------------------------------
 # fused GCBF+ loss (K10)
 # --------------------------------------------------------------------------

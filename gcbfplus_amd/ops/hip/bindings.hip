@@ -25,6 +25,9 @@ __global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf1
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
+__global__ void raytrace_sphere_topk_kernel(const float*, const float*, const float*, float*,
+                                            float*, bool*, int, int,
+                                            int, int, int, int, float);
 __global__ void mb_gather_kernel(const float*, const bool*, const bool*, const bool*,
                                  const float*, const long*, float*, bool*, bool*, bool*,
                                  float*, int, int, int, int);
@@ -46,9 +49,14 @@ __global__ void di_loss_prep_fwd_kernel(const float*, const float*, const float*
 __global__ void di_loss_prep_bwd_kernel(const float*, const float*, const float*, const float*,
                                         const float*, const float*, float*, int, int, int,
                                         float, float, float, float);
-__global__ void di_env_step_kernel(const float*, const float*, const float*, const float*,
-                                   float*, bool*, float*, float*, int, int, int, float,
-                                   float, float, float, float);
+template <int DYN>
+__global__ void env_step2d_kernel(const float*, const float*, const float*, const float*,
+                                  float*, bool*, float*, float*, int, int, int, float,
+                                  float, float, float, float);
+__global__ void drone3d_step_kernel(const float*, const float*, const float*, const float*,
+                                    const float*, const float*, float*, bool*, float*,
+                                    float*, int, int, int, float, float, float, float,
+                                    float);
 __global__ void gcbf_loss_bwd_kernel(const float*, const float*, const float*, const float*, const float*, const bool*, const bool*, const float*, const float*, float*, float*, float*, float*, long, int, float, float, float, float, float, float, float);
 
 #define CHECK_IN(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
@@ -344,6 +352,48 @@ torch::Tensor raytrace_rect(torch::Tensor pos, torch::Tensor points, long n_rays
   return hits;
 }
 
+torch::Tensor raytrace_sphere_topk(torch::Tensor pos, torch::Tensor centers,
+                                   torch::Tensor radii, long n_beams, long topk,
+                                   double range) {
+  CHECK_IN(pos);
+  CHECK_IN(centers);
+  CHECK_IN(radii);
+  long B = pos.size(0), N = pos.size(1), K = centers.size(1);
+  const int nt = (int)n_beams / 2;
+  const int R = nt * (int)n_beams + 2;
+  auto hits = torch::empty({B, N, topk, 3}, pos.options());
+  size_t smem = ((size_t)K * 4 + ((R + 1) & ~1)) * sizeof(float)
+                + 256 * sizeof(unsigned long long);
+  hipLaunchKernelGGL(raytrace_sphere_topk_kernel, dim3(B * N), dim3(256), smem, cur_stream(),
+                     pos.data_ptr<float>(), centers.data_ptr<float>(),
+                     radii.data_ptr<float>(), hits.data_ptr<float>(),
+                     (float*)nullptr, (bool*)nullptr, 0, 0,
+                     (int)N, (int)K, (int)n_beams, (int)topk, (float)range);
+  return hits;
+}
+
+void raytrace_sphere_graph(torch::Tensor pos, torch::Tensor centers, torch::Tensor radii,
+                           torch::Tensor states_out, torch::Tensor mask_out,
+                           long n_beams, long topk, double range) {
+  CHECK_IN(pos);
+  CHECK_IN(centers);
+  CHECK_IN(radii);
+  CHECK_IN(states_out);
+  CHECK_IN(mask_out);
+  long B = pos.size(0), N = pos.size(1), K = centers.size(1);
+  long S = states_out.size(2), D = mask_out.size(2);
+  const int nt = (int)n_beams / 2;
+  const int R = nt * (int)n_beams + 2;
+  size_t smem = ((size_t)K * 4 + ((R + 1) & ~1)) * sizeof(float)
+                + 256 * sizeof(unsigned long long);
+  hipLaunchKernelGGL(raytrace_sphere_topk_kernel, dim3(B * N), dim3(256), smem, cur_stream(),
+                     pos.data_ptr<float>(), centers.data_ptr<float>(),
+                     radii.data_ptr<float>(), (float*)nullptr,
+                     states_out.data_ptr<float>(), mask_out.data_ptr<bool>(),
+                     (int)S, (int)D,
+                     (int)N, (int)K, (int)n_beams, (int)topk, (float)range);
+}
+
 torch::Tensor fused_adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                                torch::Tensor v, torch::Tensor pbf, torch::Tensor t,
                                double lr, double b1, double b2, double eps, double wd,
@@ -490,7 +540,7 @@ std::vector<torch::Tensor> gcbf_loss_bwd(torch::Tensor h, torch::Tensor h_next,
 std::vector<torch::Tensor> di_env_step(torch::Tensor states, torch::Tensor action,
                                        torch::Tensor points, torch::Tensor Kmat, long N,
                                        long R, double dt, double inv_m, double comm,
-                                       double car_r, double vmax) {
+                                       double car_r, double vmax, long dyn) {
   CHECK_IN(states);
   CHECK_IN(action);
   CHECK_IN(points);
@@ -502,12 +552,50 @@ std::vector<torch::Tensor> di_env_step(torch::Tensor states, torch::Tensor actio
   auto reward = torch::empty({B}, states.options());
   auto cost = torch::empty({B}, states.options());
   size_t smem = (size_t)(N * 4 + N * 2 + K * 8) * sizeof(float);
-  hipLaunchKernelGGL(di_env_step_kernel, dim3(B), dim3(256), smem, cur_stream(),
+  if (dyn == 0)
+    hipLaunchKernelGGL(env_step2d_kernel<0>, dim3(B), dim3(256), smem, cur_stream(),
+                       states.data_ptr<float>(), action.data_ptr<float>(),
+                       points.data_ptr<float>(), Kmat.data_ptr<float>(),
+                       nxt.data_ptr<float>(), mask.data_ptr<bool>(),
+                       reward.data_ptr<float>(), cost.data_ptr<float>(), (int)N, (int)K,
+                       (int)R, (float)dt, (float)inv_m, (float)comm, (float)car_r,
+                       (float)vmax);
+  else
+    hipLaunchKernelGGL(env_step2d_kernel<1>, dim3(B), dim3(256), smem, cur_stream(),
+                       states.data_ptr<float>(), action.data_ptr<float>(),
+                       points.data_ptr<float>(), Kmat.data_ptr<float>(),
+                       nxt.data_ptr<float>(), mask.data_ptr<bool>(),
+                       reward.data_ptr<float>(), cost.data_ptr<float>(), (int)N, (int)K,
+                       (int)R, (float)dt, (float)inv_m, (float)comm, (float)car_r,
+                       (float)vmax);
+  return {nxt, mask, reward, cost};
+}
+
+std::vector<torch::Tensor> drone3d_step(torch::Tensor states, torch::Tensor action,
+                                        torch::Tensor centers, torch::Tensor radii,
+                                        torch::Tensor Kmat, torch::Tensor Amat, long N,
+                                        long R, double dt, double bgain, double comm,
+                                        double drone_r, double vmax) {
+  CHECK_IN(states);
+  CHECK_IN(action);
+  CHECK_IN(centers);
+  CHECK_IN(radii);
+  CHECK_IN(Kmat);
+  CHECK_IN(Amat);
+  long B = states.size(0), K = centers.size(1);
+  long D = N + 1 + R;
+  auto nxt = torch::empty_like(states);
+  auto mask = torch::empty({B, N, D}, states.options().dtype(torch::kBool));
+  auto reward = torch::empty({B}, states.options());
+  auto cost = torch::empty({B}, states.options());
+  size_t smem = (size_t)(N * 6 + N * 3 + K * 4) * sizeof(float);
+  hipLaunchKernelGGL(drone3d_step_kernel, dim3(B), dim3(256), smem, cur_stream(),
                      states.data_ptr<float>(), action.data_ptr<float>(),
-                     points.data_ptr<float>(), Kmat.data_ptr<float>(),
+                     centers.data_ptr<float>(), radii.data_ptr<float>(),
+                     Kmat.data_ptr<float>(), Amat.data_ptr<float>(),
                      nxt.data_ptr<float>(), mask.data_ptr<bool>(),
                      reward.data_ptr<float>(), cost.data_ptr<float>(), (int)N, (int)K,
-                     (int)R, (float)dt, (float)inv_m, (float)comm, (float)car_r,
+                     (int)R, (float)dt, (float)bgain, (float)comm, (float)drone_r,
                      (float)vmax);
   return {nxt, mask, reward, cost};
 }
@@ -548,7 +636,13 @@ torch::Tensor di_loss_prep_bwd(torch::Tensor states, torch::Tensor raw, torch::T
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("di_loss_prep_fwd", &di_loss_prep_fwd);
   m.def("di_loss_prep_bwd", &di_loss_prep_bwd);
-  m.def("di_env_step", &di_env_step, "fused DoubleIntegrator env step (K5-K8)");
+  m.def("di_env_step", &di_env_step,
+        "fused 2D env step, dyn=0 DI / dyn=1 Dubins (K5-K8)",
+        py::arg("states"), py::arg("action"), py::arg("points"),
+        py::arg("Kmat"), py::arg("N"), py::arg("R"), py::arg("dt"),
+        py::arg("inv_m"), py::arg("comm"), py::arg("car_r"),
+        py::arg("vmax"), py::arg("dyn") = 0);
+  m.def("drone3d_step", &drone3d_step, "fused LinearDrone step part A");
   m.def("gcbf_loss_fwd", &gcbf_loss_fwd);
   m.def("gcbf_loss_bwd", &gcbf_loss_bwd);
   m.def("edge_msg_in_fwd", &edge_msg_in_fwd);
@@ -565,4 +659,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_aggr_bwd", &softmax_aggr_bwd);
   m.def("mb_gather", &mb_gather, "fused 5-tensor minibatch gather (K18)");
   m.def("raytrace_rect", &raytrace_rect);
+  m.def("raytrace_sphere_topk", &raytrace_sphere_topk);
+  m.def("raytrace_sphere_graph", &raytrace_sphere_graph);
 }

@@ -56,7 +56,13 @@ def proxqp_solve(
         return ext.proxqp_solve(f32c(H), f32c(g), f32c(C), f32c(b), f32c(l), f32c(u),
                                 iters, rho, sigma, alpha)
     k = b.shape[1]
-    dtype = torch.float32
+    # f64 on GPU: ROCm's f32 batched cholesky/cholesky_solve produced wrong
+    # factors at n~192 on trivially-conditioned systems (cond ~9; verified
+    # against the same data on CPU — see profiles/qp_f64_note.md). MI355X
+    # fp64 is fast and this torch path only serves QPs too big for the
+    # one-wave HIP kernel, so precision is free insurance.
+    dtype = torch.float64 if g.is_cuda else torch.float32
+    out_dtype = torch.float32
     H = H.to(dtype)
     g = g.to(dtype)
     C = C.to(dtype)
@@ -183,7 +189,7 @@ def proxqp_solve(
         x_best = torch.where(better[:, None], xp.to(dtype), x_best)
         s_best = torch.minimum(sp, s_best)
         x_cur = xp
-    return x_best
+    return x_best.to(out_dtype)
 
 
 def qp_kkt_residuals(H, g, C, b, l, u, x) -> Tuple[Tensor, Tensor]:

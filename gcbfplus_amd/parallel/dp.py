@@ -35,12 +35,21 @@ def world_size() -> int:
     return dist.get_world_size() if is_active() else 1
 
 
-def setup_from_env() -> int:
-    """Init process group from torchrun env vars; returns local rank."""
+def setup_from_env(backend: str = None) -> int:
+    """Init process group from torchrun env vars; returns local rank.
+
+    Backend: nccl(=RCCL) when every rank can own its own GPU; gloo otherwise
+    (CPU runs, or over-subscribed validation runs where ranks share one GPU —
+    RCCL refuses two ranks on the same device: 'Duplicate GPU detected')."""
     if "WORLD_SIZE" not in os.environ or int(os.environ["WORLD_SIZE"]) <= 1:
         return 0
+    world = int(os.environ["WORLD_SIZE"])
     if not dist.is_initialized():
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if backend is None:
+            backend = ("nccl" if torch.cuda.is_available()
+                       and torch.cuda.device_count() >= min(
+                           world, int(os.environ.get("LOCAL_WORLD_SIZE", world)))
+                       else "gloo")
         dist.init_process_group(backend=backend)
     local = int(os.environ.get("LOCAL_RANK", 0))
     if torch.cuda.is_available():
@@ -50,12 +59,23 @@ def setup_from_env() -> int:
     return local
 
 
+def _gloo_cuda(t: torch.Tensor) -> bool:
+    """True when the collective must stage through host memory (gloo backend
+    with device-resident tensors)."""
+    return t.is_cuda and dist.get_backend() == "gloo"
+
+
 def broadcast_modules(modules: Iterable[torch.nn.Module]):
     if not is_active():
         return
     for m in modules:
         for p in m.parameters():
-            dist.broadcast(p.data, src=0)
+            if _gloo_cuda(p.data):
+                host = p.data.cpu()
+                dist.broadcast(host, src=0)
+                p.data.copy_(host)
+            else:
+                dist.broadcast(p.data, src=0)
 
 
 @torch.no_grad()
@@ -67,7 +87,12 @@ def allreduce_mean_grads(params: List[torch.nn.Parameter]):
     if not grads:
         return
     flat = torch.cat([g.reshape(-1) for g in grads])
-    dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+    if _gloo_cuda(flat):
+        host = flat.cpu()
+        dist.all_reduce(host, op=dist.ReduceOp.SUM)
+        flat = host.to(flat.device)
+    else:
+        dist.all_reduce(flat, op=dist.ReduceOp.SUM)
     flat /= world_size()
     off = 0
     for g in grads:
@@ -90,10 +115,19 @@ def all_agree(flag: bool) -> bool:
 
 @torch.no_grad()
 def allreduce_mean_flat(flats: List[torch.Tensor]):
-    """Mean all-reduce of pre-flattened grad buckets (one call per bucket)."""
+    """Mean all-reduce of pre-flattened grad buckets (one call per bucket;
+    callers fuse everything into a single bucket — see algo/gcbf.py
+    dp_gbuf). Gloo + CUDA tensors stage through pinned host copies."""
     if not is_active():
         return
     w = world_size()
     for f in flats:
-        dist.all_reduce(f, op=dist.ReduceOp.SUM)
+        if _gloo_cuda(f):
+            host = f.cpu()
+            dist.all_reduce(host, op=dist.ReduceOp.SUM)
+            f.copy_(host)
+        else:
+            dist.all_reduce(f, op=dist.ReduceOp.SUM)
         f /= w
+
+

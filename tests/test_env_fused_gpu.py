@@ -1,0 +1,141 @@
+"""GPU oracle tests for the round-2 fused env kernels: the 3D sphere
+raytrace + top-k, the DubinsCar fused step (env_step2d_kernel<1>) and the
+LinearDrone two-kernel step — each against the composed-torch path of the
+same math (CPU fp32 or eager GPU via GCBF_NO_FUSED_ENV)."""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from gcbfplus_amd import ops
+
+
+def test_raytrace_sphere_topk_matches_cpu():
+    from gcbfplus_amd.env import make_env
+    from gcbfplus_amd.env.utils import get_lidar
+    from gcbfplus_amd.env.obstacle import Sphere
+
+    env = make_env("LinearDrone", num_agents=6, area_size=3.0, max_step=4,
+                   device="cuda")
+    rng = np.random.default_rng(70)
+    obs = env.sample_obstacles(4, rng)  # CPU tensors
+    pos = torch.rand(4, 6, 3) * 3.0
+    # force one origin inside a sphere (alpha=0 path)
+    pos[0, 0] = obs.center[0, 0]
+    hits_gpu = ops.raytrace_sphere_topk(
+        pos.cuda(), obs.center.cuda(), obs.radius.cuda(),
+        env._params["n_rays"], env.N_HIT_RETURNS,
+        env._params["comm_radius"]).cpu()
+    hits_cpu = get_lidar(pos, obs, env._params["n_rays"],
+                         env._params["comm_radius"],
+                         max_returns=env.N_HIT_RETURNS)
+    # same beams selected when alphas are distinct; compare as sets to be
+    # robust to fp ties: every GPU hit must be near some CPU hit & vice versa
+    d = torch.cdist(hits_gpu.reshape(24, 16, 3), hits_cpu.reshape(24, 16, 3))
+    # rows (gpu->cpu) and cols (cpu->gpu) nearest-match; scale tolerance for
+    # the 1e6-alpha far points (coordinates ~5e5)
+    scale = hits_cpu.reshape(24, 16, 3).abs().amax(-1).clamp_min(1.0)
+    assert (d.min(dim=2).values / scale).max() < 1e-3
+    assert (d.min(dim=1).values / scale).max() < 1e-3
+
+
+def test_dubins_fused_step_matches_eager():
+    from gcbfplus_amd.env import make_env
+
+    torch.manual_seed(71)
+    env = make_env("DubinsCar", num_agents=8, area_size=4.0, max_step=8,
+                   device="cuda")
+    rng = np.random.default_rng(71)
+    g = env.reset(4, rng)
+    # random states incl. nonzero v / theta, actions beyond clip range
+    st = g.states.clone()
+    st[:, :8, 3] = torch.rand(4, 8, device="cuda") * 1.6 - 0.8
+    g = env.get_graph(st[:, :8], g.states[:, 8:16], g.env_states)
+    a = torch.randn(4, 8, 2, device="cuda") * 2.5
+
+    res_f = env._step_fused(g, a)
+    os.environ["GCBF_NO_FUSED_ENV"] = "1"
+    try:
+        res_e = env.step(g, a)
+    finally:
+        os.environ.pop("GCBF_NO_FUSED_ENV")
+
+    assert torch.allclose(res_f.reward, res_e.reward, atol=1e-4), \
+        (res_f.reward - res_e.reward).abs().max()
+    assert torch.allclose(res_f.cost, res_e.cost, atol=1e-5)
+    n = 8 * 2  # agent+goal rows
+    s_err = (res_f.graph.states[:, :n] - res_e.graph.states[:, :n]).abs().max()
+    assert s_err < 1e-5, s_err
+    hits_f, hits_r = res_f.graph.states[:, n:], res_e.graph.states[:, n:]
+    close = hits_r.abs().amax(-1, keepdim=True) < 100.0
+    assert ((hits_f - hits_r).abs() * close).max() < 1e-3
+    mism = (res_f.graph.mask != res_e.graph.mask).float().mean()
+    assert mism < 1e-3, mism
+
+
+def test_drone_fused_step_matches_eager():
+    from gcbfplus_amd.env import make_env
+
+    torch.manual_seed(72)
+    env = make_env("LinearDrone", num_agents=6, area_size=3.0, max_step=8,
+                   device="cuda")
+    rng = np.random.default_rng(72)
+    g = env.reset(4, rng)
+    st = g.states.clone()
+    st[:, :6, 3:] = torch.rand(4, 6, 3, device="cuda") - 0.5
+    g = env.get_graph(st[:, :6], g.states[:, 6:12], g.env_states)
+    a = torch.randn(4, 6, 3, device="cuda")
+
+    res_f = env._step_fused(g, a)
+    os.environ["GCBF_NO_FUSED_ENV"] = "1"
+    try:
+        res_e = env.step(g, a)
+    finally:
+        os.environ.pop("GCBF_NO_FUSED_ENV")
+
+    assert torch.allclose(res_f.reward, res_e.reward, atol=1e-4)
+    assert torch.allclose(res_f.cost, res_e.cost, atol=1e-5)
+    n = 6 * 2
+    s_err = (res_f.graph.states[:, :n] - res_e.graph.states[:, :n]).abs().max()
+    assert s_err < 1e-5, s_err
+    # lidar rows: set-compare per agent (top-k tie order can differ in fp)
+    hf = res_f.graph.states[:, n:].reshape(4 * 6, 16, 6)[..., :3]
+    hr = res_e.graph.states[:, n:].reshape(4 * 6, 16, 6)[..., :3]
+    d = torch.cdist(hf, hr)
+    scale = hr.abs().amax(-1).clamp_min(1.0)
+    assert (d.min(dim=2).values / scale).max() < 1e-3
+    assert (d.min(dim=1).values / scale).max() < 1e-3
+    agm = (res_f.graph.mask[:, :, : 6 + 1] != res_e.graph.mask[:, :, : 6 + 1])
+    assert agm.float().mean() < 1e-3
+    # lidar mask bits may differ only at comm-radius fp boundaries
+    lm = (res_f.graph.mask[:, :, 7:] != res_e.graph.mask[:, :, 7:])
+    assert lm.float().mean() < 2e-2
+
+
+def test_dubins_rollout_fused_vs_eager():
+    """8-step policy-free rollout through both paths stays close (drift from
+    fused fp ordering must be tiny)."""
+    from gcbfplus_amd.env import make_env
+
+    env = make_env("DubinsCar", num_agents=8, area_size=4.0, max_step=8,
+                   device="cuda")
+    rng = np.random.default_rng(73)
+    g0 = env.reset(2, rng)
+
+    def roll(fused: bool):
+        if not fused:
+            os.environ["GCBF_NO_FUSED_ENV"] = "1"
+        try:
+            g = g0
+            for _ in range(8):
+                a = env.u_ref(g)
+                g = env.step(g, a).graph
+            return g.states[:, :16].clone()
+        finally:
+            os.environ.pop("GCBF_NO_FUSED_ENV", None)
+
+    s_f, s_e = roll(True), roll(False)
+    assert (s_f - s_e).abs().max() < 1e-3, (s_f - s_e).abs().max()
