@@ -86,7 +86,10 @@ def get_lidar(
     hits = starts + (ends - starts) * alphas[..., None]
     hits = hits.reshape(B, N, R, dim)
     if max_returns < R:
-        idx = torch.argsort(alphas.reshape(B, N, R), dim=-1)[:, :, :max_returns]
+        # stable: the reference's jnp.argsort keeps tie order (all the
+        # alpha=1e6 no-hit beams), so the selected beam INDICES matter, not
+        # just the sorted distances — torch default argsort is unstable
+        idx = torch.argsort(alphas.reshape(B, N, R), dim=-1, stable=True)[:, :, :max_returns]
         hits = torch.gather(hits, 2, idx[..., None].expand(B, N, max_returns, dim))
     return hits
 

@@ -163,6 +163,81 @@ class _FusedLinearHIP(torch.autograd.Function):
         return dx, dw, db, None
 
 
+class _FusedLinearOneHotHIP(torch.autograd.Function):
+    """y = act(x @ w[oh:] + (b + w[oh-1])) for the agents-only GNN update
+    layer whose first `oh` input features are the constant agent one-hot
+    [0,0,1]. Replaces the eager `w[3:]` slice + `b + w[2]` add whose autograd
+    backward costs ~10 small kernels per call (zeros-pad of the sliced dW,
+    scatter, AccumulateGrad adds): here dW accumulates straight into
+    w.grad[oh:] and db into BOTH b.grad and w.grad[oh-1] inside the dW
+    reduction (gemm_tn_acc2)."""
+
+    @staticmethod
+    def forward(ctx, x: Tensor, w: Tensor, b: Tensor, act: int, oh: int):
+        ext = _require_ext()
+        x_bf = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
+        w_bf = getattr(w, "_bf", None)
+        if w_bf is None:
+            w_bf = w.to(torch.bfloat16)
+        w_bf3 = w_bf[oh:].contiguous() if not w_bf[oh:].is_contiguous() else w_bf[oh:]
+        bias_eff = b + w[oh - 1]
+        y = ext.gemm_bias_act(x_bf.contiguous(), w_bf3, bias_eff.contiguous(), act)
+        ctx.save_for_backward(x_bf, w_bf3, y)
+        ctx.act = act
+        ctx.oh = oh
+        ctx.x_dtype = x.dtype
+        ctx.direct = (w.is_leaf and b.is_leaf and w.grad is not None
+                      and b.grad is not None and w.grad.is_contiguous()
+                      and b.grad.is_contiguous())
+        ctx.params = (w, b) if ctx.direct else None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: Tensor):
+        ext = _require_ext()
+        x_bf, w_bf3, y = ctx.saved_tensors
+        oh = ctx.oh
+        if y.dtype == torch.float32:
+            if ctx.act == ACT_RELU:
+                dz_f = dy * (y > 0)
+            elif ctx.act == ACT_TANH:
+                dz_f = dy * (1.0 - y * y)
+            else:
+                dz_f = dy
+            dz = dz_f.to(torch.bfloat16).contiguous()
+            actin, yact = ACT_NONE, dz
+        else:
+            dz = dy.contiguous().to(torch.bfloat16)
+            actin, yact = ctx.act, y
+            if actin != ACT_NONE and dz.shape[1] % 32 != 0:
+                dz = ext.act_bwd(dz, y, actin)
+                actin, yact = ACT_NONE, dz
+        dx = dw_full = db_out = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.gemm_bt(dz, w_bf3, yact, actin).to(ctx.x_dtype)
+        if ctx.direct:
+            w, b = ctx.params
+            ext.gemm_tn_acc2(x_bf, dz, yact, actin, w.grad[oh:], b.grad,
+                             w.grad[oh - 1])
+        else:
+            dw, db = ext.gemm_tn(x_bf, dz, yact, actin)
+            dw_full = torch.zeros(oh + dw.shape[0], dw.shape[1],
+                                  device=dw.device, dtype=dw.dtype)
+            dw_full[oh:] = dw
+            dw_full[oh - 1] = db
+            db_out = db
+        return dx, dw_full, db_out, None, None
+
+
+def fused_linear_onehot(x: Tensor, w: Tensor, b: Tensor, act: int, oh: int = 3) -> Tensor:
+    """GPU-only one-hot fold (see _FusedLinearOneHotHIP). Callers fall back
+    to fused_linear(x, w[oh:], b + w[oh-1], act) on CPU."""
+    lead = x.shape[:-1]
+    x2 = x.reshape(-1, x.shape[-1])
+    y = _FusedLinearOneHotHIP.apply(x2, w, b, act, oh)
+    return y.reshape(*lead, w.shape[1])
+
+
 def fused_linear(x: Tensor, w: Tensor, b: Optional[Tensor], act: int = ACT_NONE) -> Tensor:
     """act(x @ w + b). x: (..., K); w: (K, N) fp32 master weight; b: (N,) fp32.
 

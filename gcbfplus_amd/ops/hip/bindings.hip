@@ -14,7 +14,7 @@ template <int ACT>
 __global__ void gemm_bias_act_bn128_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
 template <int ACT, bool BT>
 __global__ void gemm_bias_act_glds_kernel(const bf16_t_*, const bf16_t_*, const float*, bf16_t_*, int, int, int);
-__global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, long, int, int, int);
+__global__ void reduce_dw_db_kernel(const float*, const float*, float*, float*, float*, long, int, int, int);
 template <int ACT, typename OutT>
 __global__ void gemv_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, OutT*, int, int, int);
 template <int ACT>
@@ -22,6 +22,10 @@ __global__ void dot_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
 template <int ACT>
 __global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+template <int ACT>
+__global__ void gemm_tn_partial2_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+template <int ACT>
+__global__ void gemm_tn_partial3_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
@@ -184,17 +188,31 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, long act) {
 
 std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
                                         torch::Tensor yact, long actin,
-                                        torch::Tensor dw, torch::Tensor db, bool acc) {
+                                        torch::Tensor dw, torch::Tensor db, bool acc,
+                                        c10::optional<torch::Tensor> db2 = c10::nullopt) {
   CHECK_IN(x);
   CHECK_IN(dz);
   long M = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == M);
-  long gk = (K + 63) / 64, gn = (N + 63) / 64;
+  // kernel variants: 1 = 64x64 transposed-X stage (scalar LDS writes),
+  // 2 = 128x128 tile, 3 = dW^T orientation (vector-only staging). Default
+  // experiment switch GCBF_TN_KERNEL; shape-deterministic per process.
+  static int variant = [] {
+    const char* e = getenv("GCBF_TN_KERNEL");
+    return e ? atoi(e) : 3;
+  }();
+  bool big = variant == 2 && (K >= 128) && (N >= 128);
+  long tk = big ? 128 : 64, tn = big ? 128 : 64;
+  long gk = (K + tk - 1) / tk, gn = (N + tn - 1) / tn;
   // deterministic split count: aim for ~1024 blocks, depends on shapes only
   long S = std::min<long>(128, std::max<long>(1, 1024 / std::max<long>(1, gk * gn)));
   // keep >= 128 rows per split so small-M dW calls don't pay 64x partial
   // traffic (S still a pure function of shapes: deterministic)
   S = std::min<long>(S, std::max<long>(1, (M + 127) / 128));
+  // XCD-aware 1-D launch requires S % 8 == 0 (extra slabs see empty row
+  // ranges and write zero partials — harmless, still deterministic)
+  bool remap = S >= 8;
+  if (remap) S = (S + 7) / 8 * 8;
 
   auto opts = x.options().dtype(torch::kFloat32);
   auto partial = torch::empty({S, K, N}, opts);
@@ -202,16 +220,27 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   auto stream = cur_stream();
   const bf16_t_* ya = actin ? bfp(yact) : nullptr;
   auto launch = [&](auto kernel) {
-    hipLaunchKernelGGL(kernel, dim3(gk, gn, S), dim3(256), 0, stream,
-                       bfp(x), bfp(dz), ya, partial.data_ptr<float>(),
+    hipLaunchKernelGGL(kernel, remap ? dim3(gk * gn * S) : dim3(gk, gn, S), dim3(256), 0,
+                       stream, bfp(x), bfp(dz), ya, partial.data_ptr<float>(),
                        db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
   };
-  if (actin == 1) launch(gemm_tn_partial_kernel<1>);
-  else if (actin == 2) launch(gemm_tn_partial_kernel<2>);
-  else launch(gemm_tn_partial_kernel<0>);
+  if (big) {
+    if (actin == 1) launch(gemm_tn_partial2_kernel<1>);
+    else if (actin == 2) launch(gemm_tn_partial2_kernel<2>);
+    else launch(gemm_tn_partial2_kernel<0>);
+  } else if (variant == 3) {
+    if (actin == 1) launch(gemm_tn_partial3_kernel<1>);
+    else if (actin == 2) launch(gemm_tn_partial3_kernel<2>);
+    else launch(gemm_tn_partial3_kernel<0>);
+  } else {
+    if (actin == 1) launch(gemm_tn_partial_kernel<1>);
+    else if (actin == 2) launch(gemm_tn_partial_kernel<2>);
+    else launch(gemm_tn_partial_kernel<0>);
+  }
   hipLaunchKernelGGL(reduce_dw_db_kernel, dim3((K * N + N + 255) / 256), dim3(256), 0, stream,
                      partial.data_ptr<float>(), db_partial.data_ptr<float>(),
-                     dw.data_ptr<float>(), db.data_ptr<float>(), K * N, (int)N, (int)S,
+                     dw.data_ptr<float>(), db.data_ptr<float>(),
+                     db2 ? db2->data_ptr<float>() : nullptr, K * N, (int)N, (int)S,
                      acc ? 1 : 0);
   return {dw, db};
 }
@@ -229,6 +258,16 @@ void gemm_tn_acc(torch::Tensor x, torch::Tensor dz, torch::Tensor yact, long act
   TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() && db.is_contiguous());
   TORCH_CHECK(dw.size(0) == x.size(1) && dw.size(1) == dz.size(1) && db.size(0) == dz.size(1));
   gemm_tn_impl(x, dz, yact, actin, dw, db, true);
+}
+
+// one-hot fold backward: dw += X^T dZ into a kernel.grad row-slice, db +=
+// into BOTH bias.grad and kernel.grad[oh_row] in the same reduction pass
+void gemm_tn_acc2(torch::Tensor x, torch::Tensor dz, torch::Tensor yact, long actin,
+                  torch::Tensor dw, torch::Tensor db, torch::Tensor db2) {
+  TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() && db.is_contiguous() && db2.is_contiguous());
+  TORCH_CHECK(dw.size(0) == x.size(1) && dw.size(1) == dz.size(1) && db.size(0) == dz.size(1));
+  TORCH_CHECK(db2.numel() == db.numel());
+  gemm_tn_impl(x, dz, yact, actin, dw, db, true, db2);
 }
 
 
@@ -643,6 +682,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("inv_m"), py::arg("comm"), py::arg("car_r"),
         py::arg("vmax"), py::arg("dyn") = 0);
   m.def("drone3d_step", &drone3d_step, "fused LinearDrone step part A");
+  m.def("gemm_tn_acc2", &gemm_tn_acc2);
   m.def("gcbf_loss_fwd", &gcbf_loss_fwd);
   m.def("gcbf_loss_bwd", &gcbf_loss_bwd);
   m.def("edge_msg_in_fwd", &edge_msg_in_fwd);

@@ -435,9 +435,27 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
   const int lane = tid & 63;
   const int w = tid >> 6;
   const int wk = w >> 1, wn = w & 1;  // wave tile 32(K) x 32(N)
-  const int k0 = blockIdx.x * BKDIM;
-  const int n0 = blockIdx.y * BN;
-  const int s = blockIdx.z;
+  // XCD-aware placement (1-D launch, S % 8 == 0): all gk*gn blocks of one
+  // split-M slab s get ids with id%8 == s%8 -> same XCD (dispatch places
+  // block b on XCD b%8), so the X and dZ slab reads of the 2nd..24th block
+  // hit that XCD's L2 instead of re-reading HBM from 6 different XCDs.
+  int kb, nb, s;
+  if (gridDim.y == 1) {
+    const int gk = (K + BKDIM - 1) / BKDIM, gn = (N + BN - 1) / BN;
+    const int per = gk * gn;
+    const int id = blockIdx.x;
+    const int rest = id >> 3;
+    s = (id & 7) + 8 * (rest / per);
+    const int j = rest % per;
+    kb = j / gn;
+    nb = j % gn;
+  } else {
+    kb = blockIdx.x;
+    nb = blockIdx.y;
+    s = blockIdx.z;
+  }
+  const int k0 = kb * BKDIM;
+  const int n0 = nb * BN;
 
   const long m_per = ((long)M + S - 1) / S;
   const long ms = (long)s * m_per;
@@ -511,7 +529,7 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
     __syncthreads();
     if (m0 + BMR < me) load_tile(m0 + BMR);  // overlap with the MFMA below
 
-    if (blockIdx.x == 0) {
+    if (kb == 0) {
 #pragma unroll
       for (int q = 0; q < 2; ++q) {
         const bf16x8 v = *(const bf16x8*)(&sB[db_q * 2 + q][db_c][0]);
@@ -536,7 +554,7 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
     }
   }
 
-  if (blockIdx.x == 0) {
+  if (kb == 0) {
     // db_acc covers rows {db_q*16..+16} interleaved (q pairs): tree-reduce
     sDb[db_q][db_c] = db_acc;
     __syncthreads();
@@ -563,12 +581,327 @@ template __global__ void gemm_tn_partial_kernel<0>(const bf16_t*, const bf16_t*,
 template __global__ void gemm_tn_partial_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
 template __global__ void gemm_tn_partial_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
 
+// ---------------------------------------------------------------------------
+// dW computed TRANSPOSED: dW^T[n,k] = sum_m dZ[m,n] X[m,k]. With i=n, j=k,
+// p=m BOTH MFMA operands are m-blocked LDS images ([m/8][col][8]) staged
+// with single b128 vector writes — no scalar transpose stores at all (the
+// 64x64 kernel above spends its time on 8-way scalar LDS writes for the
+// transposed X stage). Global loads are column-wise b16 but lane-coalesced
+// (adjacent lanes take adjacent columns). The 16x16 C fragments are written
+// back transposed so the partial keeps the (S, K, N) layout.
+// Block 64(n) x 64(k), 4 waves 2x2 of 32x32, BMR=64 with register prefetch.
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
+                             const bf16_t* __restrict__ Yact,
+                             float* __restrict__ partial, float* __restrict__ db_partial,
+                             int M, int N, int K, int S) {
+  constexpr int BNR = 64, BKD = 64, BMR = 64;
+  __shared__ bf16_t sZ[BMR / 8][BNR][8];  // dZ image (A-operand)
+  __shared__ bf16_t sX[BMR / 8][BKD][8];  // X image (B-operand)
+  __shared__ float sDb[4][BNR];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wi = w >> 1, wj = w & 1;  // wave tile 32(n) x 32(k)
+  int kb, nb, s;
+  if (gridDim.y == 1) {
+    const int gk = (K + BKD - 1) / BKD, gn = (N + BNR - 1) / BNR;
+    const int per = gk * gn;
+    const int id = blockIdx.x;
+    const int rest = id >> 3;
+    s = (id & 7) + 8 * (rest / per);
+    const int j = rest % per;
+    kb = j / gn;
+    nb = j % gn;
+  } else {
+    kb = blockIdx.x;
+    nb = blockIdx.y;
+    s = blockIdx.z;
+  }
+  const int k0 = kb * BKD;
+  const int n0 = nb * BNR;
+
+  const long m_per = ((long)M + S - 1) / S;
+  const long ms = (long)s * m_per;
+  const long me = (ms + m_per < (long)M) ? ms + m_per : (long)M;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float db_acc = 0.f;
+  const int db_c = tid & 63, db_q = tid >> 6;
+
+  // staging: 512 col-chunks each (64 cols x 8 m-blocks); 2 per thread.
+  //   chunk c: col = c & 63, m-block = c >> 6. Column-wise global b16 loads
+  //   (lane-coalesced in col), one b128 LDS store into the image.
+  bf16_t xv[2][8], zv[2][8];
+
+  auto load_tile = [&](long m0) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int c = tid + h * 256;
+      const int col = c & 63;
+      const long mr0 = m0 + (c >> 6) * 8;
+      const bool kin = k0 + col < K;
+      const bool nin = n0 + col < N;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const long mr = mr0 + i;
+        xv[h][i] = (mr < me && kin) ? X[mr * K + k0 + col] : (bf16_t)0.f;
+        float z = (mr < me && nin) ? (float)dZ[mr * N + n0 + col] : 0.f;
+        if constexpr (ACT != 0)
+          if (mr < me && nin) z *= act_grad_from_out((float)Yact[mr * N + n0 + col], ACT);
+        zv[h][i] = (bf16_t)z;
+      }
+    }
+  };
+
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int c = tid + h * 256;
+      const int col = c & 63;
+      const int mb = c >> 6;
+      *(bf16x8*)(&sX[mb][col][0]) = *(bf16x8*)xv[h];
+      *(bf16x8*)(&sZ[mb][col][0]) = *(bf16x8*)zv[h];
+    }
+  };
+
+  load_tile(ms);
+  for (long m0 = ms; m0 < me; m0 += BMR) {
+    __syncthreads();
+    write_tile();
+    __syncthreads();
+    if (m0 + BMR < me) load_tile(m0 + BMR);  // overlap with the MFMA below
+
+    if (kb == 0) {
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        const bf16x8 v = *(const bf16x8*)(&sZ[db_q * 2 + q][db_c][0]);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
+      }
+    }
+#pragma unroll
+    for (int mm = 0; mm < 2; ++mm) {  // two 32-deep reduction substeps
+      bf16x8 afr[2], bfr[2];
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        afr[nf] = *(const bf16x8*)(&sZ[mm * 4 + (lane >> 4)][wi * 32 + nf * 16 + (lane & 15)][0]);
+#pragma unroll
+      for (int kf = 0; kf < 2; ++kf)
+        bfr[kf] = *(const bf16x8*)(&sX[mm * 4 + (lane >> 4)][wj * 32 + kf * 16 + (lane & 15)][0]);
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+        for (int kf = 0; kf < 2; ++kf)
+          acc[nf][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[nf], bfr[kf], acc[nf][kf], 0, 0, 0);
+    }
+  }
+
+  if (kb == 0) {
+    sDb[db_q][db_c] = db_acc;
+    __syncthreads();
+    if (db_q == 0 && n0 + db_c < N)
+      db_partial[(long)s * N + n0 + db_c] =
+          (sDb[0][db_c] + sDb[1][db_c]) + (sDb[2][db_c] + sDb[3][db_c]);
+  }
+
+  // C frag: lane holds C[row = n-frag][col = k-frag]; write transposed into
+  // partial[s][k][n] (4-float runs are contiguous in n? no: row=n varies by
+  // r -> stride 1 writes along n for fixed k = per-lane column)
+  float* out = partial + (long)s * K * N;
+#pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+    for (int kf = 0; kf < 2; ++kf) {
+      const int krow = k0 + wj * 32 + kf * 16 + (lane & 15);
+      if (krow >= K) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int ncol = n0 + wi * 32 + nf * 16 + (lane >> 4) * 4 + r;
+        if (ncol < N) out[(long)krow * N + ncol] = acc[nf][kf][r];
+      }
+    }
+  }
+}
+
+template __global__ void gemm_tn_partial3_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+
+// ---------------------------------------------------------------------------
+// dW = X^T dZ, 128x128 output tile / 64x64 wave tile (K >= 128, N >= 128).
+// The 64x64 kernel above is LDS-load bound (1 b128 fragment read per MFMA,
+// measured 121 TF); the 64x64 WAVE tile reuses each fragment 4x -> 0.5
+// loads/MFMA. Same split-M partial + fused-db scheme, same XCD-aware 1-D
+// placement decode.
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void gemm_tn_partial2_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
+                             const bf16_t* __restrict__ Yact,
+                             float* __restrict__ partial, float* __restrict__ db_partial,
+                             int M, int N, int K, int S) {
+  constexpr int BK2 = 128, BN2 = 128, BMR = 64, TPAD = 72;
+  __shared__ bf16_t sXT[BK2][TPAD];      // [k][m] transposed X tile (18.4 KB)
+  __shared__ bf16_t sB[BMR / 8][BN2][8]; // dZ tile, m-blocked (16.4 KB)
+  __shared__ float sDb[2][BN2];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wk = w >> 1, wn = w & 1;  // wave tile 64(K) x 64(N)
+  int kb, nb, s;
+  if (gridDim.y == 1) {
+    const int gk = (K + BK2 - 1) / BK2, gn = (N + BN2 - 1) / BN2;
+    const int per = gk * gn;
+    const int id = blockIdx.x;
+    const int rest = id >> 3;
+    s = (id & 7) + 8 * (rest / per);
+    const int j = rest % per;
+    kb = j / gn;
+    nb = j % gn;
+  } else {
+    kb = blockIdx.x;
+    nb = blockIdx.y;
+    s = blockIdx.z;
+  }
+  const int k0 = kb * BK2;
+  const int n0 = nb * BN2;
+
+  const long m_per = ((long)M + S - 1) / S;
+  const long ms = (long)s * m_per;
+  const long me = (ms + m_per < (long)M) ? ms + m_per : (long)M;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float db_acc = 0.f;
+  const int db_c = tid & 127, db_h = tid >> 7;
+
+  // staging: 1024 8-elem chunks each for X and dZ -> 4 chunks per thread
+  //   chunk c: local row mr = c >> 4 (64 rows), 8-col group (c & 15) * 8
+  bf16_t xv[4][8], zv[4][8];
+
+  auto load_tile = [&](long m0) {
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
+      const int c = tid + h * 256;
+      const long mr = m0 + (c >> 4);
+      const int kc = (c & 15) * 8;
+      if (mr < me && k0 + kc + 7 < K) {
+        *(bf16x8*)xv[h] = *(const bf16x8*)(X + mr * K + k0 + kc);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          xv[h][i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
+      }
+      if (mr < me && n0 + kc + 7 < N) {
+        *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + mr * N + n0 + kc);
+        if constexpr (ACT != 0) {
+          bf16x8 yv = *(const bf16x8*)(Yact + mr * N + n0 + kc);
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            zv[h][i] = (bf16_t)((float)zv[h][i] * act_grad_from_out((float)yv[i], ACT));
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          const bool ok = mr < me && n0 + kc + i < N;
+          float z = ok ? (float)dZ[mr * N + n0 + kc + i] : 0.f;
+          if constexpr (ACT != 0)
+            if (ok) z *= act_grad_from_out((float)Yact[mr * N + n0 + kc + i], ACT);
+          zv[h][i] = (bf16_t)z;
+        }
+      }
+    }
+  };
+
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
+      const int c = tid + h * 256;
+      const int mr = c >> 4;
+      const int kc = (c & 15) * 8;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) sXT[kc + i][mr] = xv[h][i];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) sB[mr >> 3][kc + i][mr & 7] = zv[h][i];
+    }
+  };
+
+  load_tile(ms);
+  for (long m0 = ms; m0 < me; m0 += BMR) {
+    __syncthreads();
+    write_tile();
+    __syncthreads();
+    if (m0 + BMR < me) load_tile(m0 + BMR);  // overlap with the MFMA below
+
+    if (kb == 0) {
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const bf16x8 v = *(const bf16x8*)(&sB[db_h * 4 + q][db_c][0]);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
+      }
+    }
+#pragma unroll
+    for (int mm = 0; mm < 2; ++mm) {  // two 32-deep reduction substeps
+      bf16x8 afr[4], bfr[4];
+#pragma unroll
+      for (int kf = 0; kf < 4; ++kf)
+        afr[kf] = *(const bf16x8*)(&sXT[wk * 64 + kf * 16 + (lane & 15)][mm * 32 + (lane >> 4) * 8]);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        bfr[nf] = *(const bf16x8*)(&sB[mm * 4 + (lane >> 4)][wn * 64 + nf * 16 + (lane & 15)][0]);
+#pragma unroll
+      for (int kf = 0; kf < 4; ++kf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[kf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[kf], bfr[nf], acc[kf][nf], 0, 0, 0);
+    }
+  }
+
+  if (kb == 0) {
+    sDb[db_h][db_c] = db_acc;
+    __syncthreads();
+    if (db_h == 0 && n0 + db_c < N)
+      db_partial[(long)s * N + n0 + db_c] = sDb[0][db_c] + sDb[1][db_c];
+  }
+
+  float* out = partial + (long)s * K * N;
+#pragma unroll
+  for (int kf = 0; kf < 4; ++kf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int col = n0 + wn * 64 + nf * 16 + (lane & 15);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int krow = k0 + wk * 64 + kf * 16 + (lane >> 4) * 4 + r;
+        if (krow < K) out[(long)krow * N + col] = acc[kf][nf][r];
+      }
+    }
+  }
+}
+
+template __global__ void gemm_tn_partial2_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial2_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial2_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+
 // (dw_partial (S,K,N), db_partial (S,N)) -> (dW, db) in ONE launch
 // (fixed-order sums: deterministic; 4 accumulators hide add latency).
 // acc != 0 accumulates (+=) into dW/db — used to write straight into the
 // optimizer's flat-gradient views, skipping autograd's AccumulateGrad adds.
 __global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* __restrict__ pb,
                                     float* __restrict__ dW, float* __restrict__ db,
+                                    float* __restrict__ db2,
                                     long KN, int N, int S, int acc) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const bool is_db = i >= KN;
@@ -588,6 +921,9 @@ __global__ void reduce_dw_db_kernel(const float* __restrict__ pw, const float* _
   const float r = (a0 + a1) + (a2 + a3);
   float* dst = is_db ? db + off : dW + off;
   *dst = acc ? (*dst + r) : r;
+  // optional second db accumulation target (one-hot bias fold: the same db
+  // flows into both bias.grad and kernel.grad[oh_row])
+  if (is_db && db2 != nullptr) db2[off] += r;
 }
 
 // ---------------------------------------------------------------------------

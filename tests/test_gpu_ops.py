@@ -600,3 +600,28 @@ def test_mb_gather_matches_index_select():
     _C.mb_gather(states, masks, safe, unsafe, u_qp, idx, *outs)
     for out, src in zip(outs, (states, masks, safe, unsafe, u_qp)):
         assert torch.equal(out, src[idx])
+
+
+def test_fused_linear_onehot_matches_composed():
+    """One-hot fold with direct grad accumulation vs the composed slice path
+    (values and parameter grads)."""
+    torch.manual_seed(80)
+    M, C, N = 512, 128, 256
+    x = torch.randn(M, C, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(3 + C, N, device="cuda", requires_grad=True)
+    b = torch.randn(N, device="cuda", requires_grad=True)
+    w.grad = torch.zeros_like(w)
+    b.grad = torch.zeros_like(b)
+    y = ops.fused_linear_onehot(x, w, b, ops.ACT_RELU)
+    g = torch.randn_like(y)
+    (y * g).sum().backward()
+    dw1, db1 = w.grad.clone(), b.grad.clone()
+
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = ops.fused_linear(x, w2[3:], b2 + w2[2], ops.ACT_RELU)
+    assert torch.equal(y, y2)
+    (y2 * g).sum().backward()
+    assert torch.allclose(db1, b2.grad, atol=1e-5), (db1 - b2.grad).abs().max()
+    assert torch.allclose(dw1, w2.grad, atol=1e-5), (dw1 - w2.grad).abs().max()
+    assert dw1[:2].abs().max().item() == 0.0
