@@ -195,12 +195,15 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   long M = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == M);
   // kernel variants: 1 = 64x64 transposed-X stage (scalar LDS writes),
-  // 2 = 128x128 tile, 3 = dW^T orientation (vector-only staging). Default
-  // experiment switch GCBF_TN_KERNEL; shape-deterministic per process.
-  static int variant = [] {
+  // 2 = 128x128 tile (slower: same scalar staging), 3 = dW^T orientation
+  // (vector-only LDS staging; measured 154 vs 121 TF on the big training
+  // shape). Shape-deterministic: 3 for large M, 1 for small/launch-bound M.
+  // GCBF_TN_KERNEL overrides for experiments.
+  static int forced = [] {
     const char* e = getenv("GCBF_TN_KERNEL");
-    return e ? atoi(e) : 3;
+    return e ? atoi(e) : 0;
   }();
+  int variant = forced ? forced : (M >= 16384 ? 3 : 1);
   bool big = variant == 2 && (K >= 128) && (N >= 128);
   long tk = big ? 128 : 64, tn = big ? 128 : 64;
   long gk = (K + tk - 1) / tk, gn = (N + tn - 1) / tn;

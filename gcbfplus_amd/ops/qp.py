@@ -101,8 +101,22 @@ def proxqp_solve(
     AtA = As.transpose(1, 2) @ As
     rho_v = torch.full((M, 1), rho, dtype=dtype, device=g.device)
 
+    on_gpu = g.is_cuda
+
     def factor(rv):
         K = Hs + sigma * eye_n + rv[:, :, None] * AtA
+        # GPU: rocSOLVER's batched factorization ops race with the current
+        # stream when free-running (per-op results verify correct with the
+        # stream drained, but the async loop diverges — see
+        # profiles/qp_f64_note.md). Keep rocSOLVER OUT of the iteration
+        # loop: factor + explicit inverse here (bracketed by syncs; factors
+        # are rare), then the loop is pure bmm.
+        if on_gpu:
+            torch.cuda.synchronize()
+            Lc = torch.linalg.cholesky(K)
+            Kinv = torch.cholesky_inverse(Lc)
+            torch.cuda.synchronize()
+            return Kinv
         return torch.linalg.cholesky(K)
 
     Lc = factor(rho_v)
@@ -112,7 +126,10 @@ def proxqp_solve(
 
     for it in range(iters):
         rhs = sigma * x - gs + torch.einsum("mkn,mk->mn", As, rho_v * z - y)
-        xt = torch.cholesky_solve(rhs.unsqueeze(-1), Lc).squeeze(-1)
+        if on_gpu:
+            xt = (Lc @ rhs.unsqueeze(-1)).squeeze(-1)  # Lc holds K^-1 here
+        else:
+            xt = torch.cholesky_solve(rhs.unsqueeze(-1), Lc).squeeze(-1)
         zt = torch.einsum("mkn,mn->mk", As, xt)
         x = alpha * xt + (1 - alpha) * x
         z_relax = alpha * zt + (1 - alpha) * z
@@ -175,7 +192,11 @@ def proxqp_solve(
         w = active.to(torch.float64)
         Kp = Hf + mu * torch.einsum("mki,mk,mkj->mij", Af, w, Af)
         rp = -gf + mu * torch.einsum("mki,mk,mk->mi", Af, w, vbound)
+        if on_gpu:  # same rocSOLVER stream-race guard as factor()
+            torch.cuda.synchronize()
         xp = torch.linalg.solve(Kp, rp)
+        if on_gpu:
+            torch.cuda.synchronize()
         # penalty multiplier estimate lambda = mu * (A xp - vbound): wrong
         # sign means the pin fights the KKT conditions -> release next pass
         resid = torch.einsum("mkn,mn->mk", Af, xp) - vbound
