@@ -143,8 +143,8 @@ class CrazyFlie(LinearDrone):
         I = torch.tensor([p["Ixx"], p["Iyy"], p["Izz"]], device=st.device)
         phi, theta, psi = st[..., PHI], st[..., THETA], st[..., PSI]
         Rm = rotmat(phi, theta, psi)
-        uvw = st[..., [U, V, W]]
-        pqr = st[..., [P_, Q_, R_]]
+        uvw = st[..., U : W + 1]  # slice: U,V,W = 6,7,8 (capture-safe)
+        pqr = st[..., R_ : P_ + 1].flip(-1)  # [P_,Q_,R_] = [11,10,9]
         v_W = torch.einsum("...ij,...j->...i", Rm, uvw)
         c_phi, s_phi = torch.cos(phi), torch.sin(phi)
         c_th = torch.cos(theta)
@@ -187,8 +187,8 @@ class CrazyFlie(LinearDrone):
     def _ll_state(self, st: Tensor) -> Tensor:
         """(phi, theta, psi, p, q, r, vx, vy, vz) world-frame (:550-562)."""
         Rm = rotmat(st[..., PHI], st[..., THETA], st[..., PSI])
-        v_W = torch.einsum("...ij,...j->...i", Rm, st[..., [U, V, W]])
-        return torch.cat([st[..., [PHI, THETA, PSI]], st[..., [P_, Q_, R_]], v_W], dim=-1)
+        v_W = torch.einsum("...ij,...j->...i", Rm, st[..., U : W + 1])
+        return torch.cat([st[..., PSI : PHI + 1].flip(-1), st[..., R_ : P_ + 1].flip(-1), v_W], dim=-1)
 
     def _ll_controls(self, st: Tensor, vel_targets: Tensor) -> Tensor:
         """Motor thrusts from the LL LQR (:564-576)."""
@@ -354,9 +354,9 @@ class CrazyFlie(LinearDrone):
         """12-dim world-frame edge state: [pos, vel_W, z-axis_W, omega_W]
         (reference edge_state, :165-182)."""
         Rm = rotmat(states[..., PHI], states[..., THETA], states[..., PSI])
-        v_W = torch.einsum("...ij,...j->...i", Rm, states[..., [U, V, W]])
+        v_W = torch.einsum("...ij,...j->...i", Rm, states[..., U : W + 1])
         z_W = Rm[..., :, 2]
-        omega_W = torch.einsum("...ij,...j->...i", Rm, states[..., [P_, Q_, R_]])
+        omega_W = torch.einsum("...ij,...j->...i", Rm, states[..., R_ : P_ + 1].flip(-1))
         return torch.cat([states[..., :3], v_W, z_W, omega_W], dim=-1)
 
     def edge_grad_to_state_jac(self, graph: GraphBatch, states: Tensor, ge: Tensor) -> Tensor:

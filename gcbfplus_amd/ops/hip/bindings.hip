@@ -26,6 +26,8 @@ template <int ACT>
 __global__ void gemm_tn_partial2_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 template <int ACT>
 __global__ void gemm_tn_partial3_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+template <int ACT>
+__global__ void gemm_tn_partial4_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
@@ -204,7 +206,7 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
     return e ? atoi(e) : 0;
   }();
   int variant = forced ? forced : (M >= 16384 ? 3 : 1);
-  bool big = variant == 2 && (K >= 128) && (N >= 128);
+  bool big = (variant == 2 || variant == 4) && (K >= 128) && (N >= 128);
   long tk = big ? 128 : 64, tn = big ? 128 : 64;
   long gk = (K + tk - 1) / tk, gn = (N + tn - 1) / tn;
   // deterministic split count: aim for ~1024 blocks, depends on shapes only
@@ -227,7 +229,11 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
                        stream, bfp(x), bfp(dz), ya, partial.data_ptr<float>(),
                        db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
   };
-  if (big) {
+  if (big && variant == 4) {
+    if (actin == 1) launch(gemm_tn_partial4_kernel<1>);
+    else if (actin == 2) launch(gemm_tn_partial4_kernel<2>);
+    else launch(gemm_tn_partial4_kernel<0>);
+  } else if (big) {
     if (actin == 1) launch(gemm_tn_partial2_kernel<1>);
     else if (actin == 2) launch(gemm_tn_partial2_kernel<2>);
     else launch(gemm_tn_partial2_kernel<0>);

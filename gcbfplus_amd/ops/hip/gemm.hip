@@ -635,9 +635,13 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   float db_acc = 0.f;
   const int db_c = tid & 63, db_q = tid >> 6;
 
-  // staging: 512 col-chunks each (64 cols x 8 m-blocks); 2 per thread.
-  //   chunk c: col = c & 63, m-block = c >> 6. Column-wise global b16 loads
-  //   (lane-coalesced in col), one b128 LDS store into the image.
+  // staging, 2 chunks per thread each:
+  //   X: column-wise (col = c & 63, m-block = c >> 6): 8 lane-coalesced b16
+  //      global loads -> ONE b128 LDS store (X has no act fold).
+  //   dZ: row-wise like the fwd kernels (row = c >> 3, 8 cols = (c&7)*8):
+  //      b128 dZ (+ b128 Yact for the act fold) -> 8 scalar LDS stores.
+  //      Column-wise dZ staging would read Yact as 8 extra b16 loads per
+  //      chunk — measured as a net step regression.
   bf16_t xv[2][8], zv[2][8];
 
   auto load_tile = [&](long m0) {
@@ -647,15 +651,30 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
       const int col = c & 63;
       const long mr0 = m0 + (c >> 6) * 8;
       const bool kin = k0 + col < K;
-      const bool nin = n0 + col < N;
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         const long mr = mr0 + i;
         xv[h][i] = (mr < me && kin) ? X[mr * K + k0 + col] : (bf16_t)0.f;
-        float z = (mr < me && nin) ? (float)dZ[mr * N + n0 + col] : 0.f;
-        if constexpr (ACT != 0)
-          if (mr < me && nin) z *= act_grad_from_out((float)Yact[mr * N + n0 + col], ACT);
-        zv[h][i] = (bf16_t)z;
+      }
+      const long zr = m0 + (c >> 3);
+      const int nc = (c & 7) * 8;
+      if (zr < me && n0 + nc + 7 < N) {
+        *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + zr * N + n0 + nc);
+        if constexpr (ACT != 0) {
+          bf16x8 yv = *(const bf16x8*)(Yact + zr * N + n0 + nc);
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            zv[h][i] = (bf16_t)((float)zv[h][i] * act_grad_from_out((float)yv[i], ACT));
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          const bool ok = zr < me && n0 + nc + i < N;
+          float z = ok ? (float)dZ[zr * N + n0 + nc + i] : 0.f;
+          if constexpr (ACT != 0)
+            if (ok) z *= act_grad_from_out((float)Yact[zr * N + n0 + nc + i], ACT);
+          zv[h][i] = (bf16_t)z;
+        }
       }
     }
   };
@@ -667,7 +686,10 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
       const int col = c & 63;
       const int mb = c >> 6;
       *(bf16x8*)(&sX[mb][col][0]) = *(bf16x8*)xv[h];
-      *(bf16x8*)(&sZ[mb][col][0]) = *(bf16x8*)zv[h];
+      const int zr = c >> 3;
+      const int nc = (c & 7) * 8;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) sZ[zr >> 3][nc + i][zr & 7] = zv[h][i];
     }
   };
 
@@ -733,6 +755,148 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
 template __global__ void gemm_tn_partial3_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
 template __global__ void gemm_tn_partial3_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
 template __global__ void gemm_tn_partial3_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+
+// ---------------------------------------------------------------------------
+// dW^T orientation at 128x128 block / 64x64 wave tile: same vector-only
+// staging as kernel3, 4x fragment reuse (0.5 LDS loads per MFMA).
+// Requires K >= 128 and N >= 128 (guards handle ragged edges).
+// ---------------------------------------------------------------------------
+template <int ACT>
+__launch_bounds__(256) __global__
+void gemm_tn_partial4_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
+                             const bf16_t* __restrict__ Yact,
+                             float* __restrict__ partial, float* __restrict__ db_partial,
+                             int M, int N, int K, int S) {
+  constexpr int BNR = 128, BKD = 128, BMR = 64;
+  __shared__ bf16_t sZ[BMR / 8][BNR][8];  // 16 KB
+  __shared__ bf16_t sX[BMR / 8][BKD][8];  // 16 KB
+  __shared__ float sDb[2][BNR];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int wi = w >> 1, wj = w & 1;  // wave tile 64(n) x 64(k)
+  int kb, nb, s;
+  if (gridDim.y == 1) {
+    const int gk = (K + BKD - 1) / BKD, gn = (N + BNR - 1) / BNR;
+    const int per = gk * gn;
+    const int id = blockIdx.x;
+    const int rest = id >> 3;
+    s = (id & 7) + 8 * (rest / per);
+    const int j = rest % per;
+    kb = j / gn;
+    nb = j % gn;
+  } else {
+    kb = blockIdx.x;
+    nb = blockIdx.y;
+    s = blockIdx.z;
+  }
+  const int k0 = kb * BKD;
+  const int n0 = nb * BNR;
+
+  const long m_per = ((long)M + S - 1) / S;
+  const long ms = (long)s * m_per;
+  const long me = (ms + m_per < (long)M) ? ms + m_per : (long)M;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float db_acc = 0.f;
+  const int db_c = tid & 127, db_h = tid >> 7;
+
+  // staging: 1024 col-chunks each (128 cols x 8 m-blocks); 4 per thread
+  bf16_t xv[4][8], zv[4][8];
+
+  auto load_tile = [&](long m0) {
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
+      const int c = tid + h * 256;
+      const int col = c & 127;
+      const long mr0 = m0 + (c >> 7) * 8;
+      const bool kin = k0 + col < K;
+      const bool nin = n0 + col < N;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const long mr = mr0 + i;
+        xv[h][i] = (mr < me && kin) ? X[mr * K + k0 + col] : (bf16_t)0.f;
+        float z = (mr < me && nin) ? (float)dZ[mr * N + n0 + col] : 0.f;
+        if constexpr (ACT != 0)
+          if (mr < me && nin) z *= act_grad_from_out((float)Yact[mr * N + n0 + col], ACT);
+        zv[h][i] = (bf16_t)z;
+      }
+    }
+  };
+
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
+      const int c = tid + h * 256;
+      const int col = c & 127;
+      const int mb = c >> 7;
+      *(bf16x8*)(&sX[mb][col][0]) = *(bf16x8*)xv[h];
+      *(bf16x8*)(&sZ[mb][col][0]) = *(bf16x8*)zv[h];
+    }
+  };
+
+  load_tile(ms);
+  for (long m0 = ms; m0 < me; m0 += BMR) {
+    __syncthreads();
+    write_tile();
+    __syncthreads();
+    if (m0 + BMR < me) load_tile(m0 + BMR);
+
+    if (kb == 0) {
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const bf16x8 v = *(const bf16x8*)(&sZ[db_h * 4 + q][db_c][0]);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) db_acc += (float)v[i];
+      }
+    }
+#pragma unroll
+    for (int mm = 0; mm < 2; ++mm) {
+      bf16x8 afr[4], bfr[4];
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        afr[nf] = *(const bf16x8*)(&sZ[mm * 4 + (lane >> 4)][wi * 64 + nf * 16 + (lane & 15)][0]);
+#pragma unroll
+      for (int kf = 0; kf < 4; ++kf)
+        bfr[kf] = *(const bf16x8*)(&sX[mm * 4 + (lane >> 4)][wj * 64 + kf * 16 + (lane & 15)][0]);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+        for (int kf = 0; kf < 4; ++kf)
+          acc[nf][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[nf], bfr[kf], acc[nf][kf], 0, 0, 0);
+    }
+  }
+
+  if (kb == 0) {
+    sDb[db_h][db_c] = db_acc;
+    __syncthreads();
+    if (db_h == 0 && n0 + db_c < N)
+      db_partial[(long)s * N + n0 + db_c] = sDb[0][db_c] + sDb[1][db_c];
+  }
+
+  float* out = partial + (long)s * K * N;
+#pragma unroll
+  for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+    for (int kf = 0; kf < 4; ++kf) {
+      const int krow = k0 + wj * 64 + kf * 16 + (lane & 15);
+      if (krow >= K) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int ncol = n0 + wi * 64 + nf * 16 + (lane >> 4) * 4 + r;
+        if (ncol < N) out[(long)krow * N + ncol] = acc[nf][kf][r];
+      }
+    }
+  }
+}
+
+template __global__ void gemm_tn_partial4_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial4_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial4_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
 
 // ---------------------------------------------------------------------------
 // dW = X^T dZ, 128x128 output tile / 64x64 wave tile (K >= 128, N >= 128).

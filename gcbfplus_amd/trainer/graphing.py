@@ -120,7 +120,9 @@ class GraphedMinibatchStep:
 
 class GraphedRolloutStep:
     """Captures act_fn(graph) + env.step(graph, action) once; per-step replay
-    copies the previous outputs into the static inputs."""
+    copies the previous outputs into the static inputs. If capture fails
+    (an op in some env's step is capture-illegal), ``broken`` is set and
+    callers run the eager path instead."""
 
     def __init__(self, env, act_fn):
         self.env = env
@@ -128,6 +130,7 @@ class GraphedRolloutStep:
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.static_in: Optional[GraphBatch] = None
         self.outs = None
+        self.broken = False
 
     def _body(self):
         with torch.no_grad():
@@ -162,6 +165,19 @@ class GraphedRolloutStep:
             self.graph = _capture(self._body)
         self.graph.replay()
         return self.outs
+
+    def try_capture(self) -> bool:
+        """Capture once (warmup included); False + broken on failure."""
+        if self.broken:
+            return False
+        if self.graph is not None:
+            return True
+        try:
+            self.graph = _capture(self._body)
+            return True
+        except Exception:
+            self.broken = True
+            return False
 
     def advance(self):
         """Copy step outputs back into the static inputs for the next step."""

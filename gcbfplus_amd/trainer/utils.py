@@ -35,6 +35,10 @@ def collect_rollout(env, act_fn: Callable, graph0: GraphBatch, graphed=None) -> 
 
     if graphed is not None and graph0.states.is_cuda:
         graphed.reset(graph0)
+        if not graphed.try_capture():
+            graphed = None  # capture-illegal env step: eager fallback
+
+    if graphed is not None and graph0.states.is_cuda:
         for t in range(T):
             states[:, t] = graphed.static_in.states
             masks[:, t] = graphed.static_in.mask
