@@ -12,6 +12,7 @@ Differences vs GCBF (reference lines cited):
 from __future__ import annotations
 
 import copy
+import os
 from typing import Optional, Tuple
 
 import numpy as np
@@ -228,18 +229,40 @@ class GCBFPlus(GCBF):
             n_agents=g.n_agents, n_rays=g.n_rays,
         )
         e_big, mi_big = self._net_inputs(big)
-        h_both = self.cbf(big, e_big, msg_in=mi_big).squeeze(-1)  # (2B, N)
-        h = h_both[:B].reshape(-1)
-        h_next = h_both[B:].reshape(-1)
+        gated = (g.states.is_cuda and ops.hip_available() and self.gnn_layers == 1
+                 and not os.environ.get("GCBF_NO_GATED_NG"))
+        if gated:
+            # The stop-gradient CBF evaluation has IDENTICAL VALUES to h_next
+            # (same params — stop_gradient only cuts the param backward). So:
+            # ONE batched forward over [g; next_g], where the next_g half's
+            # parameter gradients are row-gated to labeled agents only
+            # (ops.fused_linear row_gate). The actor gradient (through
+            # next_g's states) flows for ALL rows, exactly as the reference's
+            # two-evaluation scheme (gcbf_plus.py:398-408) — and one full
+            # B-sized CBF forward+backward per minibatch disappears.
+            N = self.n_agents
+            labeled = (mb.safe | mb.unsafe)  # (B, N)
+            wgate = torch.cat([torch.ones_like(labeled), labeled], dim=0)
+            h_both = self.cbf(big, e_big, msg_in=mi_big,
+                              row_gate=wgate).squeeze(-1)  # (2B, N)
+            h = h_both[:B].reshape(-1)
+            h_next = h_both[B:].reshape(-1)
+            h_next_ng = h_next  # same tensor: the loss kernel's h_ng cotangent
+            # (unlabeled rows) and h_next cotangent (labeled rows) are
+            # row-disjoint and autograd sums them into one backward
+        else:
+            h_both = self.cbf(big, e_big, msg_in=mi_big).squeeze(-1)  # (2B, N)
+            h = h_both[:B].reshape(-1)
+            h_next = h_both[B:].reshape(-1)
 
-        # stop-gradient branch: CBF params detached, actor path alive; the
-        # VALUE equals h_next (same params) but gradients route differently
-        det_params = {k: v.detach() for k, v in self.cbf.named_parameters()}
-        e2 = None if e_big is None else e_big[B:]
-        mi2 = None if mi_big is None else mi_big[B:]
-        h_next_ng = functional_call(
-            self.cbf, det_params, (next_g, e2), {"msg_in": mi2}
-        ).squeeze(-1).reshape(-1)
+            # stop-gradient branch: CBF params detached, actor path alive; the
+            # VALUE equals h_next (same params) but gradients route differently
+            det_params = {k: v.detach() for k, v in self.cbf.named_parameters()}
+            e2 = None if e_big is None else e_big[B:]
+            mi2 = None if mi_big is None else mi_big[B:]
+            h_next_ng = functional_call(
+                self.cbf, det_params, (next_g, e2), {"msg_in": mi2}
+            ).squeeze(-1).reshape(-1)
 
         if g.states.is_cuda and ops.hip_available():
             # fused loss kernel (K10): identical math, 2 kernels vs ~160

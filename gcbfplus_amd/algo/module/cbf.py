@@ -24,7 +24,10 @@ class CBFNet(nn.Module):
         self.head = MLP(128, (256, 256), act="relu", act_final=False)
         self.out = Dense(256, 1, act="tanh")
 
-    def forward(self, graph: GraphBatch, edge_feats: Tensor, msg_in=None) -> Tensor:
-        """-> h: (B, N, 1) in [-1, 1]."""
-        x = self.gnn(graph, edge_feats, msg_in0=msg_in)
-        return self.out(self.head(x))
+    def forward(self, graph: GraphBatch, edge_feats: Tensor, msg_in=None,
+                row_gate=None) -> Tensor:
+        """-> h: (B, N, 1) in [-1, 1]. row_gate (B, N) bool: per-agent
+        parameter stop-gradient (see ops.fused_linear)."""
+        x = self.gnn(graph, edge_feats, msg_in0=msg_in, row_gate=row_gate)
+        ag = None if row_gate is None else row_gate.reshape(-1).contiguous()
+        return self.out(self.head(x, ag), ag)

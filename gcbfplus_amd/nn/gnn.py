@@ -71,16 +71,21 @@ class GNNLayer(nn.Module):
         agents_only: bool,
         msg_in: Tensor = None,  # optional fused (B, N, D, K[pad]) input
         onehot_nodes: bool = False,  # node_feats ARE the constant one-hots
+        row_gate: Tensor = None,  # (B, N) bool param stop-grad mask
     ) -> Tensor:
         B, V, F = node_feats.shape
         N, D = mask.shape[1], mask.shape[2]
+        eg = ag = None  # edge-level / agent-level flat gates
+        if row_gate is not None:
+            eg = row_gate[:, :, None].expand(B, N, D).reshape(-1).contiguous()
+            ag = row_gate.reshape(-1).contiguous()
         if msg_in is None:
             flat_idx = send_idx.reshape(-1)  # (N*D,)
             sender = node_feats[:, flat_idx].reshape(B, N, D, F)
             recv = node_feats[:, :N, None, :].expand(B, N, D, F)
             msg_in = torch.cat([edge_feats, sender, recv], dim=-1)
-        msg = self.msg_out(self.msg_mlp(msg_in))  # (B,N,D,msg_dim)
-        gate = self.attn_out(self.attn_mlp(msg)).squeeze(-1)  # (B,N,D)
+        msg = self.msg_out(self.msg_mlp(msg_in, eg), eg)  # (B,N,D,msg_dim)
+        gate = self.attn_out(self.attn_mlp(msg, eg), eg).squeeze(-1)  # (B,N,D)
         aggr = ops.masked_softmax_aggr(gate, msg, mask)  # (B,N,msg_dim)
         d0 = self.update_mlp.layers[0]
         if agents_only and onehot_nodes and d0.in_dim == 3 + self.msg_dim:
@@ -90,13 +95,16 @@ class GNNLayer(nn.Module):
             if aggr.is_cuda:
                 # direct-grad variant: dW/db accumulate into the param grads
                 # inside the dW reduction (no slice-backward kernels)
-                h = ops.fused_linear_onehot(aggr, d0.kernel, d0.bias, d0.act)
+                h = ops.fused_linear_onehot(aggr, d0.kernel, d0.bias, d0.act,
+                                            row_gate=ag)
             else:
                 h = ops.fused_linear(aggr, d0.kernel[3:], d0.bias + d0.kernel[2],
                                      d0.act)
             for l in self.update_mlp.layers[1:]:
-                h = l(h)
-            return self.update_out(h)
+                h = l(h, ag)
+            return self.update_out(h, ag)
+        assert row_gate is None, \
+            "row_gate supports only the fused agents-only single-layer path"
         if agents_only:
             upd_in = torch.cat([node_feats[:, :N], aggr.to(node_feats.dtype)], dim=-1)
         else:
@@ -125,8 +133,8 @@ class GNN(nn.Module):
         self.out_dim = out_dim
 
     def forward(self, graph: GraphBatch, edge_feats: Optional[Tensor],
-                node_feats: Optional[Tensor] = None, msg_in0: Optional[Tensor] = None
-                ) -> Tensor:
+                node_feats: Optional[Tensor] = None, msg_in0: Optional[Tensor] = None,
+                row_gate: Optional[Tensor] = None) -> Tensor:
         B = graph.batch_size
         N, R, V = graph.n_agents, graph.n_rays, graph.n_nodes
         n_layers = len(self.layers)
@@ -141,11 +149,13 @@ class GNN(nn.Module):
             node_feats = one_hot_node_feats(B, N, R, device, torch.float32)
         send_idx = sender_index(N, R, device)
         x = node_feats
+        if row_gate is not None:
+            assert n_layers == 1, "row_gate requires gnn_layers == 1"
         for i, layer in enumerate(self.layers):
             last = i == n_layers - 1
             x = layer(x, edge_feats, graph.mask, send_idx, agents_only=last,
                       msg_in=msg_in0 if i == 0 else None,
-                      onehot_nodes=onehot and i == 0)
+                      onehot_nodes=onehot and i == 0, row_gate=row_gate)
         return x  # (B, N, out_dim)
 
 

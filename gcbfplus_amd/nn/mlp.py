@@ -45,8 +45,8 @@ class Dense(nn.Module):
         self.kernel = nn.Parameter(w)
         self.bias = nn.Parameter(torch.zeros(out_dim))
 
-    def forward(self, x: Tensor) -> Tensor:
-        return ops.fused_linear(x, self.kernel, self.bias, self.act)
+    def forward(self, x: Tensor, row_gate: Optional[Tensor] = None) -> Tensor:
+        return ops.fused_linear(x, self.kernel, self.bias, self.act, row_gate)
 
     def extra_repr(self) -> str:
         return f"in={self.in_dim}, out={self.out_dim}, act={self.act}"
@@ -79,7 +79,7 @@ class MLP(nn.Module):
         self.layers = nn.ModuleList(layers)
         self.out_dim = d
 
-    def forward(self, x: Tensor) -> Tensor:
+    def forward(self, x: Tensor, row_gate: Optional[Tensor] = None) -> Tensor:
         for l in self.layers:
-            x = l(x)
+            x = l(x, row_gate)
         return x

@@ -94,7 +94,8 @@ class _FusedLinearHIP(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x: Tensor, w: Tensor, b: Optional[Tensor], act: int):
+    def forward(ctx, x: Tensor, w: Tensor, b: Optional[Tensor], act: int,
+                row_gate: Optional[Tensor] = None):
         ext = _require_ext()
         x_bf = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
         # FusedAdamW maintains a bf16 shadow per param (p._bf, updated inside
@@ -117,6 +118,9 @@ class _FusedLinearHIP(torch.autograd.Function):
                 and w.grad is not None and b.grad is not None
                 and w.grad.is_contiguous() and b.grad.is_contiguous()):
             ctx.acc = (w, b)
+        # per-row dW/db stop-gradient mask (dX unaffected): GCBF+ unlabeled
+        # h_dot rows — rows with gate=False contribute nothing to the params
+        ctx.row_gate = row_gate
         return y
 
     @staticmethod
@@ -153,14 +157,15 @@ class _FusedLinearHIP(torch.autograd.Function):
                 dx = ext.gemm_bias_act(dz, wt, _zero_bias(wt.shape[1], wt.device), ACT_NONE)
             dx = dx.to(ctx.x_dtype)
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[2]:
+            rg = ctx.row_gate
             if ctx.acc is not None:
                 wp, bp = ctx.acc
-                ext.gemm_tn_acc(x_bf, dz, yact, actin, wp.grad, bp.grad)  # += into views
+                ext.gemm_tn_acc(x_bf, dz, yact, actin, wp.grad, bp.grad, rg)
             else:
-                dw, db = ext.gemm_tn(x_bf, dz, yact, actin)  # f32 (K,N), (N,)
+                dw, db = ext.gemm_tn(x_bf, dz, yact, actin, rg)  # f32 (K,N), (N,)
         if not ctx.has_bias:
             db = None
-        return dx, dw, db, None
+        return dx, dw, db, None, None
 
 
 class _FusedLinearOneHotHIP(torch.autograd.Function):
@@ -173,7 +178,8 @@ class _FusedLinearOneHotHIP(torch.autograd.Function):
     reduction (gemm_tn_acc2)."""
 
     @staticmethod
-    def forward(ctx, x: Tensor, w: Tensor, b: Tensor, act: int, oh: int):
+    def forward(ctx, x: Tensor, w: Tensor, b: Tensor, act: int, oh: int,
+                row_gate: Optional[Tensor] = None):
         ext = _require_ext()
         x_bf = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
         w_bf = getattr(w, "_bf", None)
@@ -190,6 +196,7 @@ class _FusedLinearOneHotHIP(torch.autograd.Function):
                       and b.grad is not None and w.grad.is_contiguous()
                       and b.grad.is_contiguous())
         ctx.params = (w, b) if ctx.direct else None
+        ctx.row_gate = row_gate
         return y
 
     @staticmethod
@@ -218,30 +225,34 @@ class _FusedLinearOneHotHIP(torch.autograd.Function):
         if ctx.direct:
             w, b = ctx.params
             ext.gemm_tn_acc2(x_bf, dz, yact, actin, w.grad[oh:], b.grad,
-                             w.grad[oh - 1])
+                             w.grad[oh - 1], ctx.row_gate)
         else:
-            dw, db = ext.gemm_tn(x_bf, dz, yact, actin)
+            dw, db = ext.gemm_tn(x_bf, dz, yact, actin, ctx.row_gate)
             dw_full = torch.zeros(oh + dw.shape[0], dw.shape[1],
                                   device=dw.device, dtype=dw.dtype)
             dw_full[oh:] = dw
             dw_full[oh - 1] = db
             db_out = db
-        return dx, dw_full, db_out, None, None
+        return dx, dw_full, db_out, None, None, None
 
 
-def fused_linear_onehot(x: Tensor, w: Tensor, b: Tensor, act: int, oh: int = 3) -> Tensor:
+def fused_linear_onehot(x: Tensor, w: Tensor, b: Tensor, act: int, oh: int = 3,
+                        row_gate: Optional[Tensor] = None) -> Tensor:
     """GPU-only one-hot fold (see _FusedLinearOneHotHIP). Callers fall back
     to fused_linear(x, w[oh:], b + w[oh-1], act) on CPU."""
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
-    y = _FusedLinearOneHotHIP.apply(x2, w, b, act, oh)
+    y = _FusedLinearOneHotHIP.apply(x2, w, b, act, oh, row_gate)
     return y.reshape(*lead, w.shape[1])
 
 
-def fused_linear(x: Tensor, w: Tensor, b: Optional[Tensor], act: int = ACT_NONE) -> Tensor:
+def fused_linear(x: Tensor, w: Tensor, b: Optional[Tensor], act: int = ACT_NONE,
+                 row_gate: Optional[Tensor] = None) -> Tensor:
     """act(x @ w + b). x: (..., K); w: (K, N) fp32 master weight; b: (N,) fp32.
 
     GPU: bf16 MFMA kernel. CPU: fp32 torch (autograd oracle).
+    row_gate (flat rows,) bool: rows with False contribute nothing to dW/db
+    (dX unaffected) — per-sample parameter stop-gradient.
     """
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
@@ -256,8 +267,9 @@ def fused_linear(x: Tensor, w: Tensor, b: Optional[Tensor], act: int = ACT_NONE)
             # slices dW back to the master shape
             w = torch.cat([w, w.new_zeros(x2.shape[-1] - w.shape[0], w.shape[1])], dim=0)
     if x2.is_cuda:
-        y = _FusedLinearHIP.apply(x2, w, b, act)
+        y = _FusedLinearHIP.apply(x2, w, b, act, row_gate)
     else:
+        assert row_gate is None, "row_gate is a GPU-path feature"
         y = torch.addmm(b, x2, w) if b is not None else x2 @ w
         y = _apply_act(y, act)
     return y.reshape(*lead, w.shape[1])

@@ -21,11 +21,11 @@ template <int ACT>
 __global__ void dot_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, float*, long, int);
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
 template <int ACT>
-__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, const bool*, float*, float*, int, int, int, int);
 template <int ACT>
 __global__ void gemm_tn_partial2_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 template <int ACT>
-__global__ void gemm_tn_partial3_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+__global__ void gemm_tn_partial3_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, const bool*, float*, float*, int, int, int, int);
 template <int ACT>
 __global__ void gemm_tn_partial4_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
@@ -191,7 +191,8 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, long act) {
 std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
                                         torch::Tensor yact, long actin,
                                         torch::Tensor dw, torch::Tensor db, bool acc,
-                                        c10::optional<torch::Tensor> db2 = c10::nullopt) {
+                                        c10::optional<torch::Tensor> db2 = c10::nullopt,
+                                        c10::optional<torch::Tensor> rowgate = c10::nullopt) {
   CHECK_IN(x);
   CHECK_IN(dz);
   long M = x.size(0), K = x.size(1), N = dz.size(1);
@@ -206,6 +207,7 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
     return e ? atoi(e) : 0;
   }();
   int variant = forced ? forced : (M >= 16384 ? 3 : 1);
+  if (rowgate && (variant == 2 || variant == 4)) variant = 3;  // gate: 1/3 only
   bool big = (variant == 2 || variant == 4) && (K >= 128) && (N >= 128);
   long tk = big ? 128 : 64, tn = big ? 128 : 64;
   long gk = (K + tk - 1) / tk, gn = (N + tn - 1) / tn;
@@ -224,19 +226,30 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   auto db_partial = torch::empty({S, N}, opts);
   auto stream = cur_stream();
   const bf16_t_* ya = actin ? bfp(yact) : nullptr;
+  const bool* rg = nullptr;
+  if (rowgate) {
+    TORCH_CHECK(rowgate->is_contiguous() && rowgate->numel() == M
+                && rowgate->dtype() == torch::kBool);
+    rg = rowgate->data_ptr<bool>();
+  }
   auto launch = [&](auto kernel) {
+    hipLaunchKernelGGL(kernel, remap ? dim3(gk * gn * S) : dim3(gk, gn, S), dim3(256), 0,
+                       stream, bfp(x), bfp(dz), ya, rg, partial.data_ptr<float>(),
+                       db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
+  };
+  auto launch24 = [&](auto kernel) {  // kernels 2/4: no rowgate param
     hipLaunchKernelGGL(kernel, remap ? dim3(gk * gn * S) : dim3(gk, gn, S), dim3(256), 0,
                        stream, bfp(x), bfp(dz), ya, partial.data_ptr<float>(),
                        db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
   };
   if (big && variant == 4) {
-    if (actin == 1) launch(gemm_tn_partial4_kernel<1>);
-    else if (actin == 2) launch(gemm_tn_partial4_kernel<2>);
-    else launch(gemm_tn_partial4_kernel<0>);
+    if (actin == 1) launch24(gemm_tn_partial4_kernel<1>);
+    else if (actin == 2) launch24(gemm_tn_partial4_kernel<2>);
+    else launch24(gemm_tn_partial4_kernel<0>);
   } else if (big) {
-    if (actin == 1) launch(gemm_tn_partial2_kernel<1>);
-    else if (actin == 2) launch(gemm_tn_partial2_kernel<2>);
-    else launch(gemm_tn_partial2_kernel<0>);
+    if (actin == 1) launch24(gemm_tn_partial2_kernel<1>);
+    else if (actin == 2) launch24(gemm_tn_partial2_kernel<2>);
+    else launch24(gemm_tn_partial2_kernel<0>);
   } else if (variant == 3) {
     if (actin == 1) launch(gemm_tn_partial3_kernel<1>);
     else if (actin == 2) launch(gemm_tn_partial3_kernel<2>);
@@ -255,28 +268,31 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
 }
 
 std::vector<torch::Tensor> gemm_tn(torch::Tensor x, torch::Tensor dz, torch::Tensor yact,
-                                   long actin) {
+                                   long actin,
+                                   c10::optional<torch::Tensor> rowgate = c10::nullopt) {
   auto opts = x.options().dtype(torch::kFloat32);
   auto dw = torch::empty({x.size(1), dz.size(1)}, opts);
   auto db = torch::empty({dz.size(1)}, opts);
-  return gemm_tn_impl(x, dz, yact, actin, dw, db, false);
+  return gemm_tn_impl(x, dz, yact, actin, dw, db, false, c10::nullopt, rowgate);
 }
 
 void gemm_tn_acc(torch::Tensor x, torch::Tensor dz, torch::Tensor yact, long actin,
-                 torch::Tensor dw, torch::Tensor db) {
+                 torch::Tensor dw, torch::Tensor db,
+                 c10::optional<torch::Tensor> rowgate = c10::nullopt) {
   TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() && db.is_contiguous());
   TORCH_CHECK(dw.size(0) == x.size(1) && dw.size(1) == dz.size(1) && db.size(0) == dz.size(1));
-  gemm_tn_impl(x, dz, yact, actin, dw, db, true);
+  gemm_tn_impl(x, dz, yact, actin, dw, db, true, c10::nullopt, rowgate);
 }
 
 // one-hot fold backward: dw += X^T dZ into a kernel.grad row-slice, db +=
 // into BOTH bias.grad and kernel.grad[oh_row] in the same reduction pass
 void gemm_tn_acc2(torch::Tensor x, torch::Tensor dz, torch::Tensor yact, long actin,
-                  torch::Tensor dw, torch::Tensor db, torch::Tensor db2) {
+                  torch::Tensor dw, torch::Tensor db, torch::Tensor db2,
+                  c10::optional<torch::Tensor> rowgate = c10::nullopt) {
   TORCH_CHECK(dw.is_cuda() && dw.is_contiguous() && db.is_contiguous() && db2.is_contiguous());
   TORCH_CHECK(dw.size(0) == x.size(1) && dw.size(1) == dz.size(1) && db.size(0) == dz.size(1));
   TORCH_CHECK(db2.numel() == db.numel());
-  gemm_tn_impl(x, dz, yact, actin, dw, db, true, db2);
+  gemm_tn_impl(x, dz, yact, actin, dw, db, true, db2, rowgate);
 }
 
 
@@ -691,7 +707,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("inv_m"), py::arg("comm"), py::arg("car_r"),
         py::arg("vmax"), py::arg("dyn") = 0);
   m.def("drone3d_step", &drone3d_step, "fused LinearDrone step part A");
-  m.def("gemm_tn_acc2", &gemm_tn_acc2);
+  m.def("gemm_tn_acc2", &gemm_tn_acc2, py::arg("x"), py::arg("dz"), py::arg("yact"),
+        py::arg("actin"), py::arg("dw"), py::arg("db"), py::arg("db2"),
+        py::arg("rowgate") = py::none());
   m.def("gcbf_loss_fwd", &gcbf_loss_fwd);
   m.def("gcbf_loss_bwd", &gcbf_loss_bwd);
   m.def("edge_msg_in_fwd", &edge_msg_in_fwd);
@@ -702,8 +720,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bias_act", &gemm_bias_act, "Y = act(X@W + b), MFMA bf16");
   m.def("act_bwd", &act_bwd, "dZ = dY * act'(Y)");
   m.def("gemm_bt", &gemm_bt, "dX = dZ @ W^T (transposed B-stage, no copy)");
-  m.def("gemm_tn", &gemm_tn, "dW = X^T dZ, db = colsum dZ (deterministic)");
-  m.def("gemm_tn_acc", &gemm_tn_acc, "gemm_tn accumulated (+=) into given f32 buffers");
+  m.def("gemm_tn", &gemm_tn, "dW = X^T dZ, db = colsum dZ (deterministic)",
+        py::arg("x"), py::arg("dz"), py::arg("yact"), py::arg("actin"),
+        py::arg("rowgate") = py::none());
+  m.def("gemm_tn_acc", &gemm_tn_acc, py::arg("x"), py::arg("dz"), py::arg("yact"),
+        py::arg("actin"), py::arg("dw"), py::arg("db"),
+        py::arg("rowgate") = py::none());
   m.def("softmax_aggr_fwd", &softmax_aggr_fwd);
   m.def("softmax_aggr_bwd", &softmax_aggr_bwd);
   m.def("mb_gather", &mb_gather, "fused 5-tensor minibatch gather (K18)");

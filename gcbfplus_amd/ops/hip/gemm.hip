@@ -423,6 +423,11 @@ template <int ACT>
 __launch_bounds__(256) __global__
 void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
                             const bf16_t* __restrict__ Yact,
+                            const bool* __restrict__ rowgate,  // null or (M,):
+                            // rows with gate=false contribute NOTHING to
+                            // dW/db (the dX path elsewhere stays ungated) —
+                            // implements per-sample stop-gradient of the
+                            // parameters (GCBF+ unlabeled h_dot rows)
                             float* __restrict__ partial, float* __restrict__ db_partial,
                             int M, int N, int K, int S) {
   // 64x64 output tile, BMR=64 reduction steps with register-prefetch
@@ -487,7 +492,11 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
         for (int i = 0; i < 8; ++i)
           xv[h][i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
       }
-      if (mr < me && n0 + kc + 7 < N) {
+      const bool gated = rowgate != nullptr && mr < me && !rowgate[mr];
+      if (gated) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) zv[h][i] = (bf16_t)0.f;
+      } else if (mr < me && n0 + kc + 7 < N) {
         *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + mr * N + n0 + kc);
         if constexpr (ACT != 0) {
           // dZ operand is dY here: fold dZ = dY * act'(Y) into the stage
@@ -577,9 +586,9 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
     }
 }
 
-template __global__ void gemm_tn_partial_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
 
 // ---------------------------------------------------------------------------
 // dW computed TRANSPOSED: dW^T[n,k] = sum_m dZ[m,n] X[m,k]. With i=n, j=k,
@@ -595,6 +604,7 @@ template <int ACT>
 __launch_bounds__(256) __global__
 void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
                              const bf16_t* __restrict__ Yact,
+                             const bool* __restrict__ rowgate,
                              float* __restrict__ partial, float* __restrict__ db_partial,
                              int M, int N, int K, int S) {
   constexpr int BNR = 64, BKD = 64, BMR = 64;
@@ -658,7 +668,11 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
       }
       const long zr = m0 + (c >> 3);
       const int nc = (c & 7) * 8;
-      if (zr < me && n0 + nc + 7 < N) {
+      const bool zgated = rowgate != nullptr && zr < me && !rowgate[zr];
+      if (zgated) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) zv[h][i] = (bf16_t)0.f;
+      } else if (zr < me && n0 + nc + 7 < N) {
         *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + zr * N + n0 + nc);
         if constexpr (ACT != 0) {
           bf16x8 yv = *(const bf16x8*)(Yact + zr * N + n0 + nc);
@@ -752,9 +766,9 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   }
 }
 
-template __global__ void gemm_tn_partial3_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial3_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial3_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
 
 // ---------------------------------------------------------------------------
 // dW^T orientation at 128x128 block / 64x64 wave tile: same vector-only
