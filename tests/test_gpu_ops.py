@@ -665,3 +665,19 @@ def test_gated_ng_matches_functional_call():
     # kernels, identical inputs -> expect bit equality; allow tiny tolerance)
     assert torch.allclose(c1, c2, atol=5e-7), (c1 - c2).abs().max()
     assert torch.allclose(a1, a2, atol=5e-7), (a1 - a2).abs().max()
+
+
+def test_gemm_tn_rowgate_matches_masked():
+    """rowgate semantics: dW/db over gated rows only; both kernel variants."""
+    from gcbfplus_amd import _C
+
+    torch.manual_seed(110)
+    for M in (4096, 20480):  # kernel1 and kernel3 dispatch
+        x = (torch.randn(M, 64, device="cuda") * 0.5).to(torch.bfloat16)
+        dz = (torch.randn(M, 128, device="cuda") * 0.5).to(torch.bfloat16)
+        gate = torch.rand(M, device="cuda") < 0.5
+        dw, db = _C.gemm_tn(x, dz, dz, 0, gate)
+        dzm = dz * gate[:, None]
+        dw_ref, db_ref = _C.gemm_tn(x, dzm.to(torch.bfloat16), dzm, 0)
+        assert torch.allclose(dw, dw_ref, atol=1e-3), (M, (dw - dw_ref).abs().max())
+        assert torch.allclose(db, db_ref, atol=1e-3), (M, (db - db_ref).abs().max())
