@@ -92,11 +92,14 @@ class GNNLayer(nn.Module):
             # agent one-hot is [0,0,1]: cat([onehot, aggr]) @ W ==
             # aggr @ W[3:] + (b + W[2]) — drops the cat/cast AND shrinks the
             # GEMM K from 3+msg_dim (padded) to msg_dim (glds-aligned)
-            if aggr.is_cuda:
+            if aggr.is_cuda and not __import__("os").environ.get("GCBF_NO_ONEHOT_FOLD"):
                 # direct-grad variant: dW/db accumulate into the param grads
                 # inside the dW reduction (no slice-backward kernels)
                 h = ops.fused_linear_onehot(aggr, d0.kernel, d0.bias, d0.act,
                                             row_gate=ag)
+            elif aggr.is_cuda:
+                h = ops.fused_linear(aggr, d0.kernel[3:].contiguous(),
+                                     d0.bias + d0.kernel[2], d0.act, ag)
             else:
                 h = ops.fused_linear(aggr, d0.kernel[3:], d0.bias + d0.kernel[2],
                                      d0.act)
