@@ -24,15 +24,24 @@ from .data import FlatBatch
 
 
 def _capture(body) -> torch.cuda.CUDAGraph:
+    # Hard syncs around the one-time capture: on this ROCm build, deep
+    # async pipelines around capture setup intermittently produce GPU
+    # memory faults (same enqueue-race family as the rocSOLVER issue in
+    # profiles/qp_f64_note.md) — with any host sync in the window the
+    # sequence is fault-free. Capture happens once per run; the syncs are
+    # free.
+    torch.cuda.synchronize()
     s = torch.cuda.Stream()
     s.wait_stream(torch.cuda.current_stream())
     with torch.cuda.stream(s):
         for _ in range(3):  # warmup (optimizer/grad lazily-built state)
             body()
     torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
     g = torch.cuda.CUDAGraph()
     with torch.cuda.graph(g):
         body()
+    torch.cuda.synchronize()
     return g
 
 
