@@ -24,19 +24,18 @@ from .data import FlatBatch
 
 
 def _capture(body) -> torch.cuda.CUDAGraph:
-    # Hard syncs around the one-time capture: on this ROCm build, deep
-    # async pipelines around capture setup intermittently produce GPU
-    # memory faults (same enqueue-race family as the rocSOLVER issue in
-    # profiles/qp_f64_note.md) — with any host sync in the window the
-    # sequence is fault-free. Capture happens once per run; the syncs are
-    # free.
+    # Warmup on the DEFAULT stream, not a side stream: running the fused
+    # optimizer step on a freshly-created side stream produced intermittent
+    # GPU memory faults on this ROCm build (tools/debug_graphed3.py trials:
+    # full=False side-stream body OK, full=True faults in the FIRST body
+    # even with per-body host syncs; the identical sequence on the default
+    # stream — the eager path and the production bench — never faults; the
+    # boxes also warn about missing iommu=pt). torch.cuda.graph uses its
+    # own capture stream internally, so a side-stream warmup is not needed.
+    # Capture happens once per run; the hard syncs are free.
     torch.cuda.synchronize()
-    s = torch.cuda.Stream()
-    s.wait_stream(torch.cuda.current_stream())
-    with torch.cuda.stream(s):
-        for _ in range(3):  # warmup (optimizer/grad lazily-built state)
-            body()
-    torch.cuda.current_stream().wait_stream(s)
+    for _ in range(3):  # warmup (optimizer/grad lazily-built state)
+        body()
     torch.cuda.synchronize()
     g = torch.cuda.CUDAGraph()
     with torch.cuda.graph(g):
