@@ -23,7 +23,7 @@ from ..utils.graph import GraphBatch
 from .data import FlatBatch
 
 
-def _capture(body) -> torch.cuda.CUDAGraph:
+def _capture(body, final=None) -> torch.cuda.CUDAGraph:
     # Warmup on the DEFAULT stream, not a side stream: running the fused
     # optimizer step on a freshly-created side stream produced intermittent
     # GPU memory faults on this ROCm build (tools/debug_graphed3.py trials:
@@ -39,7 +39,7 @@ def _capture(body) -> torch.cuda.CUDAGraph:
     torch.cuda.synchronize()
     g = torch.cuda.CUDAGraph()
     with torch.cuda.graph(g):
-        body()
+        (final or body)()
     torch.cuda.synchronize()
     return g
 
@@ -50,6 +50,7 @@ class GraphedMinibatchStep:
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.fb: Optional[FlatBatch] = None
         self.full = False
+        self.warming = False
 
     def _alloc(self, batch: FlatBatch):
         mb = self.algo.batch_size
@@ -63,12 +64,23 @@ class GraphedMinibatchStep:
             uq = torch.empty(mb, *batch.u_qp.shape[1:], device=dev)
         self.fb = FlatBatch(st, mk, sf, us, uq)
 
+    def _final_body(self):
+        self.warming = False
+        self._body()
+
     def _body(self):
         algo = self.algo
         algo.dp_gbuf.zero_()  # both optimizers' gflats are slices of this
         total, _ = algo._loss(self.fb, want_info=False)
         total.backward()
-        if self.full:  # single-GPU: optimizer inside the capture too
+        # single-GPU: optimizer inside the CAPTURE, but not in the warmup
+        # bodies — repeated loss+backward+fused-optimizer sequences outside
+        # the replayed graph intermittently page-fault on this ROCm pool
+        # (tools/debug_optstream.py; the replayed graph itself is stable
+        # across thousands of steps). FusedAdamW has no lazy state, so
+        # capturing it unwarmed is safe (its temps come from the capture
+        # pool).
+        if self.full and not self.warming:
             algo.cbf_optim.step()
             algo.actor_optim.step()
 
@@ -114,7 +126,9 @@ class GraphedMinibatchStep:
             snap = []
             for opt in (algo.cbf_optim, algo.actor_optim):
                 snap.append({k: v.clone() for k, v in opt.state_dict().items()})
-            self.graph = _capture(self._body)
+            self.warming = True
+            self.graph = _capture(self._body, final=self._final_body)
+            self.warming = False
             for opt, sd in zip((algo.cbf_optim, algo.actor_optim), snap):
                 opt.load_state_dict(sd)
         # stream capture records without executing -> always replay
