@@ -348,8 +348,8 @@ void dot_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict_
   const long m = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (m >= M) return;
   float a0 = 0.f, a1 = 0.f;
-  const int kv = (K / 16) * 16;
-  int k = 0;
+  const int kv = ((K & 7) == 0) ? (K / 16) * 16 : 0;  // b128 path needs 16-B
+  int k = 0;                                          // aligned rows (K%8==0)
   for (; k < kv; k += 16) {  // two b128 loads in flight per iteration
     const bf16x8 x0 = *(const bf16x8*)(X + m * K + k);
     const bf16x8 x1 = *(const bf16x8*)(X + m * K + k + 8);
@@ -387,7 +387,7 @@ void gemv_bias_act_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict
   float p[16];
 #pragma unroll
   for (int n = 0; n < 16; ++n) p[n] = 0.f;
-  const int kvec = (K / 8) * 8;
+  const int kvec = ((K & 7) == 0) ? K : 0;  // b128 rows need K % 8 == 0
   for (int k8 = lane * 8; k8 < kvec; k8 += WAVE * 8) {  // b128 row loads
     const bf16x8 xv = *(const bf16x8*)(X + m * K + k8);
 #pragma unroll
@@ -485,7 +485,9 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
       const int c = tid + h * 256;
       const long mr = m0 + (c >> 3);
       const int kc = (c & 7) * 8;
-      if (mr < me && k0 + kc + 7 < K) {
+      // b128 loads need 16-B alignment: row base mr*K is aligned only when
+      // K % 8 == 0 (misaligned b128 at a segment end page-faults)
+      if (mr < me && k0 + kc + 7 < K && (K & 7) == 0) {
         *(bf16x8*)xv[h] = *(const bf16x8*)(X + mr * K + k0 + kc);
       } else {
 #pragma unroll
@@ -496,7 +498,7 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
       if (gated) {
 #pragma unroll
         for (int i = 0; i < 8; ++i) zv[h][i] = (bf16_t)0.f;
-      } else if (mr < me && n0 + kc + 7 < N) {
+      } else if (mr < me && n0 + kc + 7 < N && (N & 7) == 0) {
         *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + mr * N + n0 + kc);
         if constexpr (ACT != 0) {
           // dZ operand is dY here: fold dZ = dY * act'(Y) into the stage
@@ -672,7 +674,7 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
       if (zgated) {
 #pragma unroll
         for (int i = 0; i < 8; ++i) zv[h][i] = (bf16_t)0.f;
-      } else if (zr < me && n0 + nc + 7 < N) {
+      } else if (zr < me && n0 + nc + 7 < N && (N & 7) == 0) {
         *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + zr * N + n0 + nc);
         if constexpr (ACT != 0) {
           bf16x8 yv = *(const bf16x8*)(Yact + zr * N + n0 + nc);
@@ -973,14 +975,14 @@ void gemm_tn_partial2_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
       const int c = tid + h * 256;
       const long mr = m0 + (c >> 4);
       const int kc = (c & 15) * 8;
-      if (mr < me && k0 + kc + 7 < K) {
+      if (mr < me && k0 + kc + 7 < K && (K & 7) == 0) {
         *(bf16x8*)xv[h] = *(const bf16x8*)(X + mr * K + k0 + kc);
       } else {
 #pragma unroll
         for (int i = 0; i < 8; ++i)
           xv[h][i] = (mr < me && k0 + kc + i < K) ? X[mr * K + k0 + kc + i] : (bf16_t)0.f;
       }
-      if (mr < me && n0 + kc + 7 < N) {
+      if (mr < me && n0 + kc + 7 < N && (N & 7) == 0) {
         *(bf16x8*)zv[h] = *(const bf16x8*)(dZ + mr * N + n0 + kc);
         if constexpr (ACT != 0) {
           bf16x8 yv = *(const bf16x8*)(Yact + mr * N + n0 + kc);
