@@ -21,13 +21,13 @@ template <int ACT>
 __global__ void dot_bias_act_kernel(const bf16_t_*, const bf16_t_*, const float*, float*, long, int);
 __global__ void act_bwd_kernel(const bf16_t_*, const bf16_t_*, bf16_t_*, long, int);
 template <int ACT>
-__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, const bool*, float*, float*, int, int, int, int);
+__global__ void gemm_tn_partial_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, const bool*, float*, float*, int, int, int, int, int);
 template <int ACT>
-__global__ void gemm_tn_partial2_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+__global__ void gemm_tn_partial2_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int, int);
 template <int ACT>
-__global__ void gemm_tn_partial3_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, const bool*, float*, float*, int, int, int, int);
+__global__ void gemm_tn_partial3_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, const bool*, float*, float*, int, int, int, int, int);
 template <int ACT>
-__global__ void gemm_tn_partial4_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int);
+__global__ void gemm_tn_partial4_kernel(const bf16_t_*, const bf16_t_*, const bf16_t_*, float*, float*, int, int, int, int, int);
 __global__ void softmax_aggr_fwd_kernel(const float*, const bf16_t_*, const bool*, bf16_t_*, float*, int, int);
 __global__ void softmax_aggr_bwd_kernel(const bf16_t_*, const float*, const bf16_t_*, float*, bf16_t_*, int, int);
 __global__ void raytrace_rect_kernel(const float*, const float*, float*, int, int, int, float);
@@ -235,12 +235,14 @@ std::vector<torch::Tensor> gemm_tn_impl(torch::Tensor x, torch::Tensor dz,
   auto launch = [&](auto kernel) {
     hipLaunchKernelGGL(kernel, remap ? dim3(gk * gn * S) : dim3(gk, gn, S), dim3(256), 0,
                        stream, bfp(x), bfp(dz), ya, rg, partial.data_ptr<float>(),
-                       db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
+                       db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S,
+                       remap ? 1 : 0);
   };
   auto launch24 = [&](auto kernel) {  // kernels 2/4: no rowgate param
     hipLaunchKernelGGL(kernel, remap ? dim3(gk * gn * S) : dim3(gk, gn, S), dim3(256), 0,
                        stream, bfp(x), bfp(dz), ya, partial.data_ptr<float>(),
-                       db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S);
+                       db_partial.data_ptr<float>(), (int)M, (int)N, (int)K, (int)S,
+                       remap ? 1 : 0);
   };
   if (big && variant == 4) {
     if (actin == 1) launch24(gemm_tn_partial4_kernel<1>);

@@ -429,7 +429,7 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
                             // implements per-sample stop-gradient of the
                             // parameters (GCBF+ unlabeled h_dot rows)
                             float* __restrict__ partial, float* __restrict__ db_partial,
-                            int M, int N, int K, int S) {
+                            int M, int N, int K, int S, int remap) {
   // 64x64 output tile, BMR=64 reduction steps with register-prefetch
   // staging (load tile t+1 into registers while tile t computes).
   constexpr int BKDIM = 64, BN = 64, BMR = 64, TPAD = 72;  // pad >= BMR + 8
@@ -445,7 +445,7 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
   // block b on XCD b%8), so the X and dZ slab reads of the 2nd..24th block
   // hit that XCD's L2 instead of re-reading HBM from 6 different XCDs.
   int kb, nb, s;
-  if (gridDim.y == 1) {
+  if (remap) {
     const int gk = (K + BKDIM - 1) / BKDIM, gn = (N + BN - 1) / BN;
     const int per = gk * gn;
     const int id = blockIdx.x;
@@ -588,9 +588,9 @@ void gemm_tn_partial_kernel(const bf16_t* __restrict__ X, const bf16_t* __restri
     }
 }
 
-template __global__ void gemm_tn_partial_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int, int);
 
 // ---------------------------------------------------------------------------
 // dW computed TRANSPOSED: dW^T[n,k] = sum_m dZ[m,n] X[m,k]. With i=n, j=k,
@@ -608,7 +608,7 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
                              const bf16_t* __restrict__ Yact,
                              const bool* __restrict__ rowgate,
                              float* __restrict__ partial, float* __restrict__ db_partial,
-                             int M, int N, int K, int S) {
+                             int M, int N, int K, int S, int remap) {
   constexpr int BNR = 64, BKD = 64, BMR = 64;
   __shared__ bf16_t sZ[BMR / 8][BNR][8];  // dZ image (A-operand)
   __shared__ bf16_t sX[BMR / 8][BKD][8];  // X image (B-operand)
@@ -618,7 +618,7 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   const int w = tid >> 6;
   const int wi = w >> 1, wj = w & 1;  // wave tile 32(n) x 32(k)
   int kb, nb, s;
-  if (gridDim.y == 1) {
+  if (remap) {
     const int gk = (K + BKD - 1) / BKD, gn = (N + BNR - 1) / BNR;
     const int per = gk * gn;
     const int id = blockIdx.x;
@@ -768,9 +768,9 @@ void gemm_tn_partial3_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   }
 }
 
-template __global__ void gemm_tn_partial3_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial3_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial3_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial3_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, const bool*, float*, float*, int, int, int, int, int);
 
 // ---------------------------------------------------------------------------
 // dW^T orientation at 128x128 block / 64x64 wave tile: same vector-only
@@ -782,7 +782,7 @@ __launch_bounds__(256) __global__
 void gemm_tn_partial4_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
                              const bf16_t* __restrict__ Yact,
                              float* __restrict__ partial, float* __restrict__ db_partial,
-                             int M, int N, int K, int S) {
+                             int M, int N, int K, int S, int remap) {
   constexpr int BNR = 128, BKD = 128, BMR = 64;
   __shared__ bf16_t sZ[BMR / 8][BNR][8];  // 16 KB
   __shared__ bf16_t sX[BMR / 8][BKD][8];  // 16 KB
@@ -792,7 +792,7 @@ void gemm_tn_partial4_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   const int w = tid >> 6;
   const int wi = w >> 1, wj = w & 1;  // wave tile 64(n) x 64(k)
   int kb, nb, s;
-  if (gridDim.y == 1) {
+  if (remap) {
     const int gk = (K + BKD - 1) / BKD, gn = (N + BNR - 1) / BNR;
     const int per = gk * gn;
     const int id = blockIdx.x;
@@ -910,9 +910,9 @@ void gemm_tn_partial4_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   }
 }
 
-template __global__ void gemm_tn_partial4_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial4_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial4_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial4_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial4_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial4_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int, int);
 
 // ---------------------------------------------------------------------------
 // dW = X^T dZ, 128x128 output tile / 64x64 wave tile (K >= 128, N >= 128).
@@ -926,7 +926,7 @@ __launch_bounds__(256) __global__
 void gemm_tn_partial2_kernel(const bf16_t* __restrict__ X, const bf16_t* __restrict__ dZ,
                              const bf16_t* __restrict__ Yact,
                              float* __restrict__ partial, float* __restrict__ db_partial,
-                             int M, int N, int K, int S) {
+                             int M, int N, int K, int S, int remap) {
   constexpr int BK2 = 128, BN2 = 128, BMR = 64, TPAD = 72;
   __shared__ bf16_t sXT[BK2][TPAD];      // [k][m] transposed X tile (18.4 KB)
   __shared__ bf16_t sB[BMR / 8][BN2][8]; // dZ tile, m-blocked (16.4 KB)
@@ -936,7 +936,7 @@ void gemm_tn_partial2_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   const int w = tid >> 6;
   const int wk = w >> 1, wn = w & 1;  // wave tile 64(K) x 64(N)
   int kb, nb, s;
-  if (gridDim.y == 1) {
+  if (remap) {
     const int gk = (K + BK2 - 1) / BK2, gn = (N + BN2 - 1) / BN2;
     const int per = gk * gn;
     const int id = blockIdx.x;
@@ -1071,9 +1071,9 @@ void gemm_tn_partial2_kernel(const bf16_t* __restrict__ X, const bf16_t* __restr
   }
 }
 
-template __global__ void gemm_tn_partial2_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial2_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
-template __global__ void gemm_tn_partial2_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int);
+template __global__ void gemm_tn_partial2_kernel<0>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial2_kernel<1>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int, int);
+template __global__ void gemm_tn_partial2_kernel<2>(const bf16_t*, const bf16_t*, const bf16_t*, float*, float*, int, int, int, int, int);
 
 // (dw_partial (S,K,N), db_partial (S,N)) -> (dW, db) in ONE launch
 // (fixed-order sums: deterministic; 4 accumulators hide add latency).
