@@ -72,3 +72,23 @@ def test_reference_weights_h_sign_sanity():
                        g.env_states)
     h2 = algo.get_cbf(g2)
     assert (h2[0, :2] < 0).any(), "colliding pair should get negative h"
+
+
+@pytest.mark.parametrize("env_id,n,area", [
+    ("SingleIntegrator", 8, 4.0),
+    ("DubinsCar", 8, 4.0),
+    ("LinearDrone", 8, 4.0),
+    ("CrazyFlie", 8, 4.0),
+])
+def test_reference_weights_load_all_envs(env_id, n, area):
+    """Every reference pretrained checkpoint loads and produces finite,
+    sane outputs through this framework's forward stack (full behavioral
+    evals recorded in profiles/ref_*_eval*.log / BASELINE.md)."""
+    env, algo = _algo_with_ref_weights(env_id, n, area)
+    rng = np.random.default_rng(5)
+    g = env.reset(1, rng)
+    h = algo.get_cbf(g)
+    a = algo.act(g)
+    assert torch.isfinite(h).all() and torch.isfinite(a).all()
+    r = env.step(g, a if not isinstance(a, tuple) else a[0])
+    assert torch.isfinite(r.graph.states).all()
