@@ -9,6 +9,7 @@ as obstacles, 32-ray LiDAR.
 """
 from __future__ import annotations
 
+import os
 import math
 from typing import NamedTuple, Optional, Tuple
 
@@ -237,7 +238,7 @@ class DoubleIntegrator(MultiAgentEnv):
 
     # ---- step / forward (reference :145-181, 340-354) ----------------------
     def step(self, graph: GraphBatch, action: Tensor) -> StepResult:
-        if not __import__("os").environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is DoubleIntegrator:
+        if not os.environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is DoubleIntegrator:
             return self._step_fused(graph, action)
         agent = graph.agent_states
         goal = graph.goal_states

@@ -8,6 +8,7 @@ BASELINE config #3.
 """
 from __future__ import annotations
 
+import os
 import math
 from typing import Optional, Tuple
 
@@ -110,7 +111,7 @@ class DubinsCar(DoubleIntegrator):
         return self.clip_state(graph.agent_states + x_dot * self._dt)
 
     def step(self, graph: GraphBatch, action: Tensor) -> StepResult:
-        if not __import__("os").environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is DubinsCar and self.enable_stop:
+        if not os.environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is DubinsCar and self.enable_stop:
             return self._step_fused(graph, action)
         next_agent = self._step_states(graph, action)
         reward = -((self.clip_action(action) - self.u_ref(graph)).square().sum(-1)).mean(-1)

@@ -9,6 +9,7 @@ discretized A with the continuous B (:63-72).
 """
 from __future__ import annotations
 
+import os
 import math
 from typing import Optional, Tuple
 
@@ -147,7 +148,7 @@ class LinearDrone(DoubleIntegrator):
     # cost/aa+goal mask) + part B (raytrace_sphere_topk in graph mode: hit
     # node rows + lidar mask), replacing ~40 eager launches per step --------
     def step(self, graph: GraphBatch, action: Tensor):
-        if not __import__("os").environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is LinearDrone and ops.hip_available() \
+        if not os.environ.get("GCBF_NO_FUSED_ENV") and graph.states.is_cuda and type(self) is LinearDrone and ops.hip_available() \
                 and graph.env_states.n_obs > 0:
             return self._step_fused(graph, action)
         return super().step(graph, action)

@@ -13,6 +13,7 @@ returns type_nodes(0) anyway, gnn.py:100-104).
 """
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -92,7 +93,7 @@ class GNNLayer(nn.Module):
             # agent one-hot is [0,0,1]: cat([onehot, aggr]) @ W ==
             # aggr @ W[3:] + (b + W[2]) — drops the cat/cast AND shrinks the
             # GEMM K from 3+msg_dim (padded) to msg_dim (glds-aligned)
-            if aggr.is_cuda and not __import__("os").environ.get("GCBF_NO_ONEHOT_FOLD"):
+            if aggr.is_cuda and not os.environ.get("GCBF_NO_ONEHOT_FOLD"):
                 # direct-grad variant: dW/db accumulate into the param grads
                 # inside the dW reduction (no slice-backward kernels)
                 h = ops.fused_linear_onehot(aggr, d0.kernel, d0.bias, d0.act,
