@@ -137,13 +137,25 @@ def load_flax_pickle(path: str) -> dict:
         arr.__setstate__(arr_state)
         return np.asarray(arr)
 
+    # strict allowlist: checkpoint trees contain only dicts + numpy arrays
+    # (+ the jax array wrapper); anything else is refused rather than
+    # executed (tighter than the reference's yaml.UnsafeLoader habits)
+    _ALLOWED = {
+        ("numpy.core.multiarray", "_reconstruct"): np.core.multiarray._reconstruct,
+        ("numpy", "ndarray"): np.ndarray,
+        ("numpy", "dtype"): np.dtype,
+    }
+
     class _U(pickle.Unpickler):
         def find_class(self, module, name):
             if module.startswith("jax") and name == "_reconstruct_array":
                 return _reconstruct_np
-            if module.startswith("jax"):
-                raise pickle.UnpicklingError(f"unexpected jax global {module}.{name}")
-            return super().find_class(module, name)
+            got = _ALLOWED.get((module, name))
+            if got is None:
+                raise pickle.UnpicklingError(
+                    f"refusing to unpickle global {module}.{name} "
+                    "(checkpoint allowlist: numpy arrays only)")
+            return got
 
     with open(path, "rb") as f:
         return _U(f).load()

@@ -92,3 +92,19 @@ def test_reference_weights_load_all_envs(env_id, n, area):
     assert torch.isfinite(h).all() and torch.isfinite(a).all()
     r = env.step(g, a if not isinstance(a, tuple) else a[0])
     assert torch.isfinite(r.graph.states).all()
+
+
+def test_load_flax_pickle_rejects_arbitrary_globals(tmp_path):
+    """The checkpoint unpickler must refuse non-allowlisted globals instead
+    of executing them."""
+    import pickle
+
+    class Evil:
+        def __reduce__(self):
+            return (print, ("pwned",))
+
+    p = tmp_path / "evil.pkl"
+    with open(p, "wb") as f:
+        pickle.dump({"params": Evil()}, f)
+    with pytest.raises(Exception):
+        load_flax_pickle(str(p))
