@@ -140,8 +140,12 @@ def load_flax_pickle(path: str) -> dict:
     # strict allowlist: checkpoint trees contain only dicts + numpy arrays
     # (+ the jax array wrapper); anything else is refused rather than
     # executed (tighter than the reference's yaml.UnsafeLoader habits)
+    from numpy._core import multiarray as _np_ma  # numpy 2.x home; the
+    # pickles reference the legacy "numpy.core.multiarray" module name
+
     _ALLOWED = {
-        ("numpy.core.multiarray", "_reconstruct"): np.core.multiarray._reconstruct,
+        ("numpy.core.multiarray", "_reconstruct"): _np_ma._reconstruct,
+        ("numpy._core.multiarray", "_reconstruct"): _np_ma._reconstruct,
         ("numpy", "ndarray"): np.ndarray,
         ("numpy", "dtype"): np.dtype,
     }
