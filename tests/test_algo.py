@@ -325,3 +325,56 @@ def test_bench_contract_json(tmp_path):
         assert k in d, k
     assert d["n_gpus"] == 1 and d["steps"] == 1 and d["scaling"] == "weak"
     assert d["data"] == "synthetic" and d["higher_is_better"] is True
+
+
+def test_tanh_distribution_normalizes():
+    """log_prob of the tanh-squashed Normal integrates to ~1 over [-1, 1]
+    (interior density + the two tail point-masses at +-threshold)."""
+    import torch
+    from gcbfplus_amd.algo.module.distribution import TanhTransformedDistribution
+
+    d = TanhTransformedDistribution(torch.tensor([0.3]), torch.tensor([0.7]))
+    t = d.threshold
+    xs = torch.linspace(-t + 1e-4, t - 1e-4, 20001)
+    dens = d.log_prob(xs).exp()
+    interior = torch.trapz(dens, xs)
+    lo = d.log_prob(torch.tensor([-1.0])).exp()  # tail mass below -t
+    hi = d.log_prob(torch.tensor([1.0])).exp()   # tail mass above +t
+    total = float(interior + lo + hi)
+    assert abs(total - 1.0) < 2e-3, total
+
+
+def test_tanh_distribution_entropy_sane():
+    import torch
+    from gcbfplus_amd.algo.module.distribution import TanhTransformedDistribution
+
+    torch.manual_seed(0)
+    d = TanhTransformedDistribution(torch.zeros(4096), torch.full((4096,), 0.5))
+    ent = d.entropy().mean()
+    # MC estimate of -E[log p(x)] over samples
+    s = d.sample()
+    mc = -d.log_prob(s).mean()
+    assert abs(float(ent - mc)) < 0.05, (float(ent), float(mc))
+
+
+def test_gae_matches_loop_reference():
+    import torch
+    from gcbfplus_amd.algo.utils import gae
+
+    torch.manual_seed(1)
+    b, T = 3, 7
+    r = torch.randn(b, T)
+    v = torch.randn(b, T)
+    nv = torch.randn(b, T)
+    dn = torch.rand(b, T) < 0.2
+    adv = gae(0.99, 0.95, r, v, nv, dn)
+    # plain transcription of reference algo/utils.py:18-41
+    ref = torch.zeros(b, T)
+    for i in range(b):
+        last = 0.0
+        for t in reversed(range(T)):
+            nonterm = 1.0 - float(dn[i, t])
+            delta = float(r[i, t]) + 0.99 * float(nv[i, t]) * nonterm - float(v[i, t])
+            last = delta + 0.99 * 0.95 * nonterm * last
+            ref[i, t] = last
+    assert torch.allclose(adv, ref, atol=1e-5)
