@@ -191,8 +191,14 @@ def proxqp_solve(
         vbound = torch.where(active & torch.isfinite(vbound), vbound, zf)
         w = active.to(torch.float64)
         Kp = Hf + mu * torch.einsum("mki,mk,mkj->mij", Af, w, Af)
+        # tiny ridge: H may be PSD-singular (the dec-share QP's all-10 relax
+        # block, reference dec_share_cbf.py:124 quirk, is rank-1) and a
+        # degenerate active set then leaves Kp singular; 1e-8 is negligible
+        # against the mu=1e8 penalty terms and the score check rejects any
+        # off candidate anyway
+        Kp = Kp + 1e-8 * torch.eye(n, dtype=torch.float64, device=g.device)
         rp = -gf + mu * torch.einsum("mki,mk,mk->mi", Af, w, vbound)
-        if on_gpu:  # same rocSOLVER stream-race guard as factor()
+        if on_gpu:  # sync brackets retained (see profiles/qp_f64_note.md)
             torch.cuda.synchronize()
         xp = torch.linalg.solve(Kp, rp)
         if on_gpu:
